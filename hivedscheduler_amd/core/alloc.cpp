@@ -1,0 +1,515 @@
+// Buddy allocation, virtual->physical cell mapping, free-list split/merge,
+// VC-safety accounting and bad-cell (doomed-bad) machinery.
+// Semantics parity: pkg/algorithm/cell_allocation.go:34-315 and
+// pkg/algorithm/hived_algorithm.go:453-653, 1354-1565.
+#include "core.hpp"
+
+namespace hived {
+
+namespace {
+
+struct AllocCtx {
+  const std::set<std::string>& suggested;
+  bool ignoreSuggested;
+  std::unordered_map<VirtualCell*, PhysicalCell*>& bindings;
+};
+
+// Usable = unbound, not a known-bad single-node cell, and (unless ignoring
+// suggestions) at least one node within the suggested set. Sorted by
+// opportunistic usage ascending to minimize preemption of opportunistic pods.
+std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& candidates,
+                                                  int numNeeded, const AllocCtx& ctx) {
+  std::vector<PhysicalCell*> usable;
+  for (Cell* cc : candidates) {
+    auto* c = static_cast<PhysicalCell*>(cc);
+    if (c->virt != nullptr) continue;
+    if (c->nodes.size() == 1 && !c->healthy) continue;
+    if (!ctx.ignoreSuggested) {
+      bool anySuggested = false;
+      for (auto& n : c->nodes) {
+        if (ctx.suggested.count(n)) {
+          anySuggested = true;
+          break;
+        }
+      }
+      if (!anySuggested) continue;
+    }
+    usable.push_back(c);
+  }
+  if (static_cast<int>(usable.size()) < numNeeded) return {};
+  std::stable_sort(usable.begin(), usable.end(), [](PhysicalCell* a, PhysicalCell* b) {
+    return a->usedAt(kOpportunisticPriority) < b->usedAt(kOpportunisticPriority);
+  });
+  return usable;
+}
+
+// Backtracking bipartite match of virtual cells onto candidate physical cells,
+// recursing into children to preserve the intra-cell topology.
+bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cells,
+                               const std::vector<Cell*>& candidatesIn, const AllocCtx& ctx,
+                               bool returnPicked, std::vector<PhysicalCell*>* picked) {
+  std::vector<PhysicalCell*> candidates =
+      getUsablePhysicalCells(candidatesIn, static_cast<int>(cells.size()), ctx);
+  if (candidates.empty() && !cells.empty()) return false;
+  int n = static_cast<int>(cells.size());
+  int m = static_cast<int>(candidates.size());
+  std::vector<int> pickedIdx(n, 0);
+  std::vector<bool> used(m, false);
+  int cellIndex = 0;
+  while (cellIndex >= 0) {
+    int candidateIndex;
+    for (candidateIndex = pickedIdx[cellIndex]; candidateIndex < m; candidateIndex++) {
+      if (used[candidateIndex]) continue;
+      PhysicalCell* candidate = candidates[candidateIndex];
+      bool ok;
+      if (candidate->level == kLowestLevel) {
+        ok = true;
+        ctx.bindings[cells[cellIndex]->cell] = candidate;
+      } else {
+        std::vector<BindingVertex*> childVerts;
+        childVerts.reserve(cells[cellIndex]->children.size());
+        for (auto& ch : cells[cellIndex]->children) childVerts.push_back(ch.get());
+        ok = mapVirtualCellsToPhysical(childVerts, candidate->children, ctx, false, nullptr);
+      }
+      if (ok) {
+        pickedIdx[cellIndex] = candidateIndex;
+        used[candidateIndex] = true;
+        if (cellIndex == n - 1) {
+          if (returnPicked) {
+            picked->clear();
+            for (int i = 0; i < n; i++) picked->push_back(candidates[pickedIdx[i]]);
+          }
+          return true;
+        }
+        break;
+      }
+    }
+    if (candidateIndex == m) {
+      cellIndex--;
+      if (cellIndex >= 0) {
+        used[pickedIdx[cellIndex]] = false;
+        pickedIdx[cellIndex]++;
+      }
+    } else {
+      cellIndex++;
+      if (cellIndex < n) pickedIdx[cellIndex] = 0;
+    }
+  }
+  return false;
+}
+
+// Backtracking buddy allocation: allocate a free physical cell at the virtual
+// cell's level, splitting one higher-level free cell per level on the way
+// down. Backtracking handles cells that are bad or outside suggested nodes.
+bool buddyAlloc(BindingVertex* cell, ChainCellList& freeList, int currentLevel,
+                const AllocCtx& ctx) {
+  if (currentLevel == cell->cell->level) {
+    std::vector<PhysicalCell*> picked;
+    if (mapVirtualCellsToPhysical({cell}, freeList.at(currentLevel), ctx, true, &picked)) {
+      for (PhysicalCell* c : picked) freeList.remove(c, currentLevel);
+      return true;
+    }
+    return false;
+  }
+  std::vector<PhysicalCell*> freeCells = getUsablePhysicalCells(freeList.at(currentLevel), 1, ctx);
+  if (freeCells.empty()) return false;
+  for (PhysicalCell* c : freeCells) {
+    auto saved = freeList.at(currentLevel - 1);
+    for (Cell* child : c->children) freeList.add(child, currentLevel - 1);
+    if (buddyAlloc(cell, freeList, currentLevel - 1, ctx)) {
+      freeList.remove(c, currentLevel);
+      return true;
+    }
+    freeList.at(currentLevel - 1) = saved;
+  }
+  return false;
+}
+
+// When buddy alloc fails due to bad / non-suggested cells, split higher-level
+// free cells beyond the buddy path, but only as many as safety allows
+// (splittable = free cells minus cells the VCs still need reserved).
+bool safeRelaxedBuddyAlloc(BindingVertex* cell, ChainCellList& freeList,
+                           std::map<int, int>& freeCellNum, int currentLevel,
+                           const AllocCtx& ctx) {
+  int top = freeList.top();
+  std::map<int, int> splittableNum;
+  Cell* splittableCell = nullptr;
+  for (int i = top; i > currentLevel; i--) {
+    splittableNum[i] =
+        static_cast<int>(freeList.at(i).size()) - (freeCellNum.count(i) ? freeCellNum[i] : 0);
+    if (i < top && splittableCell != nullptr) {
+      splittableNum[i] += splittableNum[i + 1] * static_cast<int>(splittableCell->children.size());
+    }
+    if (splittableCell == nullptr && !freeList.at(i).empty()) {
+      splittableCell = freeList.at(i)[0];
+    } else if (splittableCell != nullptr) {
+      splittableCell = splittableCell->children[0];
+    }
+    if (splittableNum[i] < 0) {
+      throw HivedError::Internal("VC Safety Broken: level " + std::to_string(i) +
+                                 " cell is unsplittable, splittableNum=" +
+                                 std::to_string(splittableNum[i]));
+    }
+  }
+  for (int l = currentLevel + 1; l <= top; l++) {
+    int cellNum = std::min(static_cast<int>(freeList.at(l).size()), splittableNum[l]);
+    if (cellNum <= 0) continue;
+    std::vector<Cell*> splitList;
+    for (int i = 0; i < cellNum; i++) {
+      Cell* c = freeList.at(l)[0];
+      splitList.push_back(c);
+      freeList.remove(c, l);
+    }
+    splittableNum[l] -= cellNum;
+    for (int sl = l; sl > currentLevel; sl--) {
+      std::vector<Cell*> childrenList;
+      for (Cell* sc : splitList) {
+        for (Cell* ch : sc->children) childrenList.push_back(ch);
+      }
+      splitList = std::move(childrenList);
+    }
+    // prepend the split cells so they are tried first
+    auto& cur = freeList.at(currentLevel);
+    cur.insert(cur.begin(), splitList.begin(), splitList.end());
+    std::vector<PhysicalCell*> picked;
+    if (mapVirtualCellsToPhysical({cell}, cur, ctx, true, &picked)) {
+      for (PhysicalCell* c : picked) freeList.remove(c, currentLevel);
+      return true;
+    }
+  }
+  return false;
+}
+
+int getLowestFreeCellLevel(const ChainCellList& freeList, int l) {
+  for (; l <= freeList.top(); l++) {
+    if (!freeList.at(l).empty()) return l;
+  }
+  throw HivedError::Internal("VC Safety Broken: free cell not found even at the highest level");
+}
+
+}  // namespace
+
+bool HivedCore::mapVirtualPlacementToPhysical(
+    std::vector<BindingVertex*>& preassigned, std::vector<std::vector<BindingVertex*>>& nonPreassigned,
+    ChainCellList freeList, std::map<int, int> freeCellNum,
+    const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
+    std::unordered_map<VirtualCell*, PhysicalCell*>& bindings) {
+  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings};
+  for (BindingVertex* c : preassigned) {
+    if (!buddyAlloc(c, freeList, getLowestFreeCellLevel(freeList, c->cell->level), ctx)) {
+      if (!safeRelaxedBuddyAlloc(c, freeList, freeCellNum, c->cell->level, ctx)) {
+        return false;
+      }
+    } else {
+      freeCellNum[c->cell->level]--;
+    }
+  }
+  for (auto& cells : nonPreassigned) {
+    auto* parentVirtual = static_cast<VirtualCell*>(cells[0]->cell->parent);
+    if (parentVirtual == nullptr || parentVirtual->phys == nullptr) return false;
+    if (!mapVirtualCellsToPhysical(cells, parentVirtual->phys->children, ctx, false, nullptr)) {
+      return false;
+    }
+  }
+  return true;
+}
+
+// --- free-list split/merge -------------------------------------------------
+
+// Remove a cell from the free list, splitting unsplit ancestors on the way up.
+// Returns the highest level at which a cell was removed.
+int HivedCore::removeCellFromFreeList(PhysicalCell* c) {
+  ChainCellList& freeList = freeCellList_[c->chain];
+  bool terminate = false;
+  for (;;) {
+    int l = c->level;
+    auto* parent = static_cast<PhysicalCell*>(c->parent);
+    if (parent != nullptr) {
+      if (parent->split) {
+        terminate = true;
+      } else {
+        for (Cell* child : parent->children) freeList.add(child, l);
+        parent->split = true;
+      }
+    } else {
+      terminate = true;
+    }
+    freeList.remove(c, l);
+    if (terminate) return l;
+    c = parent;
+  }
+}
+
+// Add a cell back to the free list, merging buddies into the parent while all
+// of them are free. Returns the highest level at which a cell was added.
+int HivedCore::addCellToFreeList(PhysicalCell* c) {
+  ChainCellList& freeList = freeCellList_[c->chain];
+  bool terminate = false;
+  for (;;) {
+    int l = c->level;
+    auto* parent = static_cast<PhysicalCell*>(c->parent);
+    if (parent != nullptr) {
+      bool allBuddyFree = true;
+      for (Cell* buddy : parent->children) {
+        if (buddy != c && !freeList.contains(buddy, l)) {
+          allBuddyFree = false;
+          break;
+        }
+      }
+      if (!allBuddyFree) {
+        terminate = true;
+      } else {
+        for (Cell* buddy : parent->children) {
+          if (buddy != c) freeList.remove(buddy, l);
+        }
+        parent->split = false;
+      }
+    } else {
+      terminate = true;
+    }
+    if (terminate) {
+      freeList.add(c, l);
+      return l;
+    }
+    c = parent;
+  }
+}
+
+// --- preassigned-cell allocation + safety & doomed-bad accounting -----------
+
+std::pair<bool, std::string> HivedCore::allocatePreassignedCell(PhysicalCell* c,
+                                                                const std::string& vcn,
+                                                                bool doomedBad) {
+  bool safetyOk = true;
+  std::string reason;
+  const std::string& chain = c->chain;
+  int level = c->level;
+  vcFreeCellNum_[vcn][chain][level]--;
+  allVCFreeCellNum_[chain][level]--;
+  totalLeftCellNum_[chain][level]--;
+  int splitLevelUpTo = removeCellFromFreeList(c);
+
+  Cell* parent = c->parent;
+  for (int l = level + 1; l <= splitLevelUpTo; l++) {
+    totalLeftCellNum_[chain][l]--;
+    if (totalLeftCellNum_[chain][l] < allVCFreeCellNum_[chain][l]) {
+      safetyOk = false;
+      reason = "Adding pod would lead to broken safety: cell type " + cellTypes_[chain][l] + ", " +
+               std::to_string(totalLeftCellNum_[chain][l]) + " left, " +
+               std::to_string(allVCFreeCellNum_[chain][l]) + " free cells in all VCs";
+    }
+    if (!static_cast<PhysicalCell*>(parent)->healthy) {
+      // parent bad: healthy-free count unchanged; just drop it from bad frees
+      badFreeCells_[chain].remove(parent, l);
+    } else {
+      // parent healthy: healthy-free count decreased; maybe doom VC cells
+      tryBindDoomedBadCell(chain, l);
+    }
+    parent = parent->parent;
+  }
+  if (!c->healthy) {
+    allocateBadCell(c);
+    if (!doomedBad) tryUnbindDoomedBadCell(chain, level);
+  } else {
+    tryBindDoomedBadCell(chain, level);
+  }
+  int numToReduce = static_cast<int>(c->children.size());
+  for (int l = level - 1; l >= kLowestLevel; l--) {
+    totalLeftCellNum_[chain][l] -= numToReduce;
+    if (totalLeftCellNum_[chain][l] < allVCFreeCellNum_[chain][l]) {
+      safetyOk = false;
+      reason = "Adding pod would lead to broken safety: cell type " + cellTypes_[chain][l] + ", " +
+               std::to_string(totalLeftCellNum_[chain][l]) + " left, " +
+               std::to_string(allVCFreeCellNum_[chain][l]) + " free cells in all VCs";
+    }
+    if (!doomedBad) tryBindDoomedBadCell(chain, l);
+    numToReduce *= static_cast<int>(fullCellList_[chain].at(l)[0]->children.size());
+  }
+  return {safetyOk, reason};
+}
+
+void HivedCore::releasePreassignedCell(PhysicalCell* c, const std::string& vcn, bool doomedBad) {
+  const std::string& chain = c->chain;
+  int level = c->level;
+  vcFreeCellNum_[vcn][chain][level]++;
+  allVCFreeCellNum_[chain][level]++;
+  totalLeftCellNum_[chain][level]++;
+  int mergeLevelUpTo = addCellToFreeList(c);
+
+  Cell* parent = c->parent;
+  for (int l = level + 1; l <= mergeLevelUpTo; l++) {
+    totalLeftCellNum_[chain][l]++;
+    if (!static_cast<PhysicalCell*>(parent)->healthy) {
+      badFreeCells_[chain].add(parent, l);
+    } else {
+      tryUnbindDoomedBadCell(chain, l);
+    }
+    parent = parent->parent;
+  }
+  if (!c->healthy) {
+    releaseBadCell(c);
+    if (!doomedBad) tryBindDoomedBadCell(chain, level);
+  } else {
+    tryUnbindDoomedBadCell(chain, level);
+  }
+  int numToAdd = static_cast<int>(c->children.size());
+  for (int l = level - 1; l >= kLowestLevel; l--) {
+    totalLeftCellNum_[chain][l] += numToAdd;
+    if (!doomedBad) tryUnbindDoomedBadCell(chain, l);
+    numToAdd *= static_cast<int>(fullCellList_[chain].at(l)[0]->children.size());
+  }
+}
+
+// A bad free cell being allocated: bind each bad child to a virtual cell so
+// the VC scheduler sees the failure.
+void HivedCore::allocateBadCell(PhysicalCell* c) {
+  if (badFreeCells_[c->chain].contains(c, c->level)) {
+    badFreeCells_[c->chain].remove(c, c->level);
+  }
+  if (c->virt == nullptr) {
+    VirtualCell* vc = getUnboundVirtualCell(static_cast<PhysicalCell*>(c->parent)->virt->children);
+    c->virt = vc;
+    vc->phys = c;
+  }
+  for (Cell* child : c->children) {
+    auto* pc = static_cast<PhysicalCell*>(child);
+    if (!pc->healthy) allocateBadCell(pc);
+  }
+}
+
+void HivedCore::releaseBadCell(PhysicalCell* c) {
+  badFreeCells_[c->chain].add(c, c->level);
+  if (VirtualCell* vc = c->virt) {
+    c->virt = nullptr;
+    vc->phys = nullptr;
+  }
+  for (Cell* child : c->children) {
+    auto* pc = static_cast<PhysicalCell*>(child);
+    if (!pc->healthy) releaseBadCell(pc);
+  }
+}
+
+// --- node / cell health ------------------------------------------------------
+
+void HivedCore::setNodeHealthy(const std::string& node, bool healthy) {
+  if (healthy) {
+    if (!badNodes_.count(node)) return;
+    badNodes_.erase(node);
+    auto it = nodeLeafCellsStorage_.find(node);
+    if (it == nodeLeafCellsStorage_.end()) return;
+    for (PhysicalCell* leaf : it->second) setHealthyCell(leaf);
+  } else {
+    if (badNodes_.count(node)) return;
+    badNodes_.insert(node);
+    auto it = nodeLeafCellsStorage_.find(node);
+    if (it == nodeLeafCellsStorage_.end()) return;
+    for (PhysicalCell* leaf : it->second) setBadCell(leaf);
+  }
+}
+
+// A cell is bad if ANY child is bad; propagate from leaf up.
+void HivedCore::setBadCell(PhysicalCell* c) {
+  if (!c->healthy) return;
+  c->healthy = false;
+  if (c->parent != nullptr) setBadCell(static_cast<PhysicalCell*>(c->parent));
+  if (inFreeCellList(c)) {
+    addBadFreeCell(c);
+  } else if (c->virt == nullptr && !c->split) {
+    // an ancestor is bound to a virtual cell; bind this bad cell too so the
+    // VC scheduler can see the failure
+    VirtualCell* vc = getUnboundVirtualCell(static_cast<PhysicalCell*>(c->parent)->virt->children);
+    c->virt = vc;
+    vc->phys = c;
+  }
+}
+
+// A cell is healthy if ALL children are healthy; propagate from leaf up.
+void HivedCore::setHealthyCell(PhysicalCell* c) {
+  if (c->healthy) return;
+  c->healthy = true;
+  if (inFreeCellList(c)) {
+    removeBadFreeCell(c);
+  } else if (VirtualCell* vc = c->virt) {
+    if (!c->pinned && c->priority < kMinGuaranteedPriority) {
+      // binding existed only because the cell was bad; undo it
+      c->virt = nullptr;
+      vc->phys = nullptr;
+      if (vc->parent == nullptr) {
+        // a preassigned cell: it must have been a doomed bad cell
+        vcDoomedBadCells_[vc->vc][c->chain].remove(c, c->level);
+        allVCDoomedBadCellNum_[c->chain][c->level]--;
+        releasePreassignedCell(c, vc->vc, true);
+      }
+    }
+  }
+  if (c->parent == nullptr) return;
+  for (Cell* buddy : c->parent->children) {
+    if (!static_cast<PhysicalCell*>(buddy)->healthy) return;
+  }
+  setHealthyCell(static_cast<PhysicalCell*>(c->parent));
+}
+
+void HivedCore::addBadFreeCell(PhysicalCell* c) {
+  const std::string& chain = c->chain;
+  int level = c->level;
+  badFreeCells_[chain].add(c, level);
+  if (allVCFreeCellNum_[chain][level] >
+      totalLeftCellNum_[chain][level] - static_cast<int>(badFreeCells_[chain].at(level).size())) {
+    tryBindDoomedBadCell(chain, level);
+  }
+}
+
+void HivedCore::removeBadFreeCell(PhysicalCell* c) {
+  badFreeCells_[c->chain].remove(c, c->level);
+  tryUnbindDoomedBadCell(c->chain, c->level);
+}
+
+// If healthy free cells < a VC's free cells at some level, some of the VC's
+// cells are inevitably bad: bind bad physical cells to free virtual cells so
+// users and the intra-VC scheduler can see them.
+void HivedCore::tryBindDoomedBadCell(const std::string& chain, int level) {
+  for (auto& [vcName, vcFreeNum] : vcFreeCellNum_) {
+    auto chainIt = vcFreeNum.find(chain);
+    if (chainIt == vcFreeNum.end()) continue;
+    while (chainIt->second[level] >
+           totalLeftCellNum_[chain][level] -
+               static_cast<int>(badFreeCells_[chain].at(level).size())) {
+      if (badFreeCells_[chain].at(level).empty()) return;
+      auto* pc = static_cast<PhysicalCell*>(badFreeCells_[chain].at(level)[0]);
+      auto& pre = vcSchedulers_[vcName].nonPinnedPreassigned;
+      auto preIt = pre.find(chain);
+      VirtualCell* vc = nullptr;
+      if (preIt != pre.end() && level <= preIt->second.top()) {
+        vc = getUnboundVirtualCell(preIt->second.at(level));
+      }
+      if (vc == nullptr) return;  // no more free preassigned cells to doom
+      pc->virt = vc;
+      vc->phys = pc;
+      vcDoomedBadCells_[vcName][chain].add(pc, level);
+      allVCDoomedBadCellNum_[chain][level]++;
+      allocatePreassignedCell(pc, vcName, true);
+    }
+  }
+}
+
+void HivedCore::tryUnbindDoomedBadCell(const std::string& chain, int level) {
+  for (auto& [vcName, vcFreeNum] : vcFreeCellNum_) {
+    auto chainIt = vcFreeNum.find(chain);
+    if (chainIt == vcFreeNum.end()) continue;
+    auto& doomed = vcDoomedBadCells_[vcName][chain];
+    if (level > doomed.top()) continue;
+    while (!doomed.at(level).empty() &&
+           chainIt->second[level] <
+               totalLeftCellNum_[chain][level] -
+                   static_cast<int>(badFreeCells_[chain].at(level).size())) {
+      auto* pc = static_cast<PhysicalCell*>(vcDoomedBadCells_[vcName][chain].at(level)[0]);
+      pc->virt->phys = nullptr;
+      pc->virt = nullptr;
+      vcDoomedBadCells_[vcName][chain].remove(pc, level);
+      allVCDoomedBadCellNum_[chain][level]--;
+      releasePreassignedCell(pc, vcName, true);
+    }
+  }
+}
+
+}  // namespace hived
