@@ -1,0 +1,165 @@
+"""Lightweight pod/node model + annotation codecs.
+
+Parity with reference pkg/internal/utils.go: pod filters (l.116-170),
+annotation (de)serialization with defaulting/validation (l.200-289), legacy
+gpu*->leafCell* conversion (l.189-197), binding-pod construction (l.172-186).
+
+Pods are plain dicts shaped like K8s Pod objects (works both with the real
+API server JSON and with in-memory simulation).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+import yaml
+
+from ..api import constants
+from ..api.types import PodBindInfo, PodSchedulingSpec, WebServerError
+
+Pod = Dict[str, Any]  # K8s-shaped pod dict
+
+
+def pod_key(pod: Pod) -> str:
+    meta = pod.get("metadata", {})
+    return f"{meta.get('namespace', 'default')}/{meta.get('name', '')}"
+
+
+def pod_uid(pod: Pod) -> str:
+    return pod.get("metadata", {}).get("uid", "") or pod_key(pod)
+
+
+def _annotations(pod: Pod) -> Dict[str, str]:
+    return pod.get("metadata", {}).get("annotations") or {}
+
+
+def is_completed(pod: Pod) -> bool:
+    phase = pod.get("status", {}).get("phase", "")
+    return phase in ("Succeeded", "Failed")
+
+
+def _containers_enabled(containers: List[dict]) -> bool:
+    for c in containers or []:
+        limits = (c.get("resources") or {}).get("limits") or {}
+        for name in (constants.ResourceNamePodSchedulingEnable,
+                     constants.LegacyResourceNamePodSchedulingEnable):
+            try:
+                if int(limits.get(name, 0)) > 0:
+                    return True
+            except (TypeError, ValueError):
+                pass
+    return False
+
+
+def is_hived_enabled(pod: Pod) -> bool:
+    spec = pod.get("spec", {})
+    return _containers_enabled(spec.get("initContainers")) or _containers_enabled(
+        spec.get("containers")
+    )
+
+
+def is_interested(pod: Pod) -> bool:
+    return not is_completed(pod) and is_hived_enabled(pod)
+
+
+def is_bound(pod: Pod) -> bool:
+    return bool(pod.get("spec", {}).get("nodeName")) and not is_completed(pod)
+
+
+def is_node_healthy(node: Dict[str, Any]) -> bool:
+    """Schedulable and NodeReady (reference pkg/internal/utils.go:160-170)."""
+    if node.get("spec", {}).get("unschedulable"):
+        return False
+    for c in node.get("status", {}).get("conditions", []) or []:
+        if c.get("type") == "Ready" and c.get("status") == "True":
+            return True
+    return False
+
+
+def _convert_old_annotation(s: str) -> str:
+    for old, new in (
+        ("gpuType", "leafCellType"),
+        ("gpuNumber", "leafCellNumber"),
+        ("gpuIsolation", "leafCellIsolation"),
+        ("physicalGpuIndices", "physicalLeafCellIndices"),
+    ):
+        s = s.replace(old, new)
+    return s
+
+
+def _get_annotation(pod: Pod, key: str, legacy_key: str) -> str:
+    ann = _annotations(pod)
+    return ann.get(key) or ann.get(legacy_key) or ""
+
+
+def extract_pod_scheduling_spec(pod: Pod) -> PodSchedulingSpec:
+    """Parse + default + validate the request annotation."""
+    err_pfx = f"Pod annotation {constants.AnnotationKeyPodSchedulingSpec}: "
+    raw = _get_annotation(pod, constants.AnnotationKeyPodSchedulingSpec,
+                          constants.LegacyAnnotationKeyPodSchedulingSpec)
+    if not raw:
+        raise WebServerError.bad_request(err_pfx + "Annotation does not exist or is empty")
+    try:
+        data = yaml.safe_load(_convert_old_annotation(raw)) or {}
+    except yaml.YAMLError as e:
+        raise WebServerError.bad_request(err_pfx + f"invalid YAML: {e}")
+    spec = PodSchedulingSpec.from_dict(data)
+    validate_pod_scheduling_spec(spec, pod_key(pod), err_pfx)
+    return spec
+
+
+def validate_pod_scheduling_spec(spec: PodSchedulingSpec, key: str, err_pfx: str = "") -> None:
+    from ..api.types import AffinityGroupMemberSpec, AffinityGroupSpec
+
+    # Defaulting: a pod without a group forms a singleton group "ns/name".
+    if spec.affinityGroup is None:
+        spec.affinityGroup = AffinityGroupSpec(
+            name=key,
+            members=[AffinityGroupMemberSpec(podNumber=1, leafCellNumber=spec.leafCellNumber)],
+        )
+    if not spec.virtualCluster:
+        raise WebServerError.bad_request(err_pfx + "VirtualCluster is empty")
+    if spec.priority < constants.OpportunisticPriority:
+        raise WebServerError.bad_request(
+            err_pfx + f"Priority is less than {constants.OpportunisticPriority}")
+    if spec.priority > constants.MaxGuaranteedPriority:
+        raise WebServerError.bad_request(
+            err_pfx + f"Priority is greater than {constants.MaxGuaranteedPriority}")
+    if spec.leafCellNumber <= 0:
+        raise WebServerError.bad_request(err_pfx + "LeafCellNumber is non-positive")
+    if not spec.affinityGroup.name:
+        raise WebServerError.bad_request(err_pfx + "AffinityGroup.Name is empty")
+    pod_in_group = False
+    for m in spec.affinityGroup.members:
+        if m.podNumber <= 0:
+            raise WebServerError.bad_request(err_pfx + "AffinityGroup.Members has non-positive PodNumber")
+        if m.leafCellNumber <= 0:
+            raise WebServerError.bad_request(
+                err_pfx + "AffinityGroup.Members has non-positive LeafCellNumber")
+        if m.leafCellNumber == spec.leafCellNumber:
+            pod_in_group = True
+    if not pod_in_group:
+        raise WebServerError.bad_request(err_pfx + "AffinityGroup.Members does not contain current Pod")
+
+
+def extract_pod_bind_info(pod: Pod) -> PodBindInfo:
+    raw = _get_annotation(pod, constants.AnnotationKeyPodBindInfo,
+                          constants.LegacyAnnotationKeyPodBindInfo)
+    if not raw:
+        raise WebServerError.bad_request(
+            f"Pod does not contain or contains empty annotation: {constants.AnnotationKeyPodBindInfo}")
+    data = yaml.safe_load(_convert_old_annotation(raw)) or {}
+    return PodBindInfo.from_dict(data)
+
+
+def new_binding_pod(pod: Pod, bind_info: PodBindInfo) -> Pod:
+    """Stamp node + decision annotations onto a (copied) pod."""
+    import copy
+
+    binding = copy.deepcopy(pod)
+    binding.setdefault("spec", {})["nodeName"] = bind_info.node
+    ann = binding.setdefault("metadata", {}).setdefault("annotations", {})
+    ann[constants.AnnotationKeyPodLeafCellIsolation] = ",".join(
+        str(i) for i in bind_info.leafCellIsolation)
+    ann[constants.AnnotationKeyPodBindInfo] = yaml.safe_dump(
+        bind_info.to_dict(), default_flow_style=False)
+    return binding
