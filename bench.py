@@ -1,0 +1,218 @@
+#!/usr/bin/env python3
+"""Flagship benchmark (driver contract).
+
+Measures the BASELINE.json headline on its named config:
+- scheduling p50 latency + VC-safety violations on config 4 (64 mixed
+  1/2/4-GPU requests, simulated 4-node x 8-MI355X cluster, in-memory
+  algorithm path), and
+- RCCL bus-bandwidth for scheduler-placed 1/2/4/8-GPU cells over xGMI when
+  GPUs are present (per-rank groups via torch.distributed; backend "nccl" IS
+  RCCL on ROCm).
+
+One "step" = one full 64-request scheduling round (schedule + optimistic
+commit + release), identical every step. `value` is the p50 latency of a
+single extender filter decision (schedule + commit) in milliseconds, max'd
+over ranks; lower is better.
+"""
+import argparse
+import json
+import os
+import random
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+
+def build_requests(seed: int = 0, count: int = 64):
+    rng = random.Random(seed)
+    vcs = ["VC1", "VC2", "VC3"]
+    reqs = []
+    for i in range(count):
+        reqs.append(
+            dict(
+                vc=rng.choice(vcs),
+                cells=rng.choice([1, 1, 2, 2, 4]),
+                priority=rng.choice([-1, -1, 0, 1]),
+            )
+        )
+    return reqs
+
+
+def make_sim():
+    from hivedscheduler_amd.sim import SimScheduler, mi355x_cluster_config
+
+    vcs = {
+        "VC1": [("MI355X-NODE", 2)],
+        "VC2": [("MI355X-NODE", 1), ("MI355X-NODE.MI355X-QUAD", 1)],
+        "VC3": [("MI355X-NODE.MI355X-QUAD", 1)],
+    }
+    return SimScheduler(mi355x_cluster_config(num_nodes=4, vcs=vcs))
+
+
+def run_round(sim, requests, latencies):
+    """One step: schedule all requests (timing each decision), then delete."""
+    bound = []
+    waited = 0
+    for i, req in enumerate(requests):
+        key = f"bench/p{i}"
+        spec = sim.pod_spec(vc=req["vc"], priority=req["priority"], leaf_cells=req["cells"])
+        t0 = time.perf_counter_ns()
+        r = sim.schedule(key, spec)
+        latencies.append((time.perf_counter_ns() - t0) / 1e6)  # ms
+        if r.kind == "bind":
+            bound.append(key)
+        else:
+            waited += 1
+    for key in bound:
+        sim.delete_pod(key)
+    return len(bound), waited
+
+
+def rccl_cell_probes(world, rank, local_rank):
+    """busbw for scheduler-style cells of size 1/2/4/8 (subset groups)."""
+    import torch
+    import torch.distributed as dist
+
+    from hivedscheduler_amd.probe import allreduce_probe
+
+    results = {}
+    sizes = [s for s in (1, 2, 4, 8) if s <= world]
+    for s in sizes:
+        group = dist.new_group(ranks=list(range(s))) if s < world else None
+        if rank < s:
+            probe = allreduce_probe(sizes_mb=(64,), iters=10, warmup=3, group=group)
+            if rank == 0:
+                results[str(s)] = round(probe["64"]["busbw_gbps"], 2)
+        dist.barrier()
+    return results
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--requests", type=int, default=64)
+    args = ap.parse_args()
+
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    has_cuda = torch.cuda.is_available()
+    dist = None
+    if world > 1:
+        import torch.distributed as dist
+
+        backend = "nccl" if has_cuda else "gloo"
+        if has_cuda:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    requests = build_requests(seed=0, count=args.requests)
+    sim = make_sim()
+
+    # warmup
+    warm_lat = []
+    for _ in range(args.warmup):
+        run_round(sim, requests, warm_lat)
+
+    def sync():
+        if dist is not None:
+            dist.barrier()
+        if has_cuda:
+            torch.cuda.synchronize()
+
+    latencies = []
+    sync()
+    t0 = time.perf_counter()
+    bound = waited = 0
+    for _ in range(args.steps):
+        b, w = run_round(sim, requests, latencies)
+        bound += b
+        waited += w
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cuda" if has_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    latencies.sort()
+    p50 = latencies[len(latencies) // 2]
+    p99 = latencies[min(len(latencies) - 1, int(len(latencies) * 0.99))]
+
+    # VC-safety check: after draining, every VC can take its full quota
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    violations = 0
+    for vc, sizes in (("VC1", [8, 8]), ("VC2", [8, 4]), ("VC3", [4])):
+        for j, cells in enumerate(sizes):
+            r = sim.schedule(f"bench/safety-{vc}-{j}", sim.pod_spec(vc=vc, priority=0,
+                                                                    leaf_cells=cells))
+            if r.kind != "bind":
+                violations += 1
+
+    busbw = {}
+    health = {}
+    if has_cuda:
+        if dist is not None:
+            busbw = rccl_cell_probes(world, rank, local_rank)
+        else:
+            from hivedscheduler_amd.probe import CellProbeRunner
+
+            runner = CellProbeRunner()
+            if runner.available():
+                probe = runner.probe_cell([0], size_mb=32, iters=5)
+                if probe.get("ok"):
+                    busbw = {"1": probe.get("busbw_gbps")}
+        if rank == 0:
+            try:
+                from hivedscheduler_amd.ops import gpu_health_report
+
+                rep = gpu_health_report(0, quick=True)
+                health = {"hbm_gbps": round(rep["hbm_gbps"], 1), "mfma_ok": rep["mfma_ok"]}
+            except Exception as e:  # health probes must not fail the bench
+                health = {"error": str(e)[:200]}
+
+    if rank == 0:
+        n_gpus = world if world > 1 else args.gpus
+        out = {
+            "metric": "scheduling p50 latency + VC-safety violations; RCCL bus-bw at 1/2/4/8-GPU cells",
+            "value": round(p50, 6),
+            "unit": "ms",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "hived-sched-config4-fragmentation-stress",
+                "cluster": "simulated 4-node x 8-MI355X",
+                "requests_per_step": args.requests,
+                "global_batch": args.requests,
+                "seq_len": 0,
+                "parallelism": f"dp{n_gpus}" if n_gpus > 1 else "single",
+                "p99_ms": round(p99, 6),
+                "binds_per_step": bound // max(1, args.steps),
+                "vc_safety_violations": violations,
+                "rccl_busbw_gbps": busbw,
+                "gpu_health": health,
+            },
+        }
+        print(json.dumps(out))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,50 @@
+"""HIP (gfx950) GPU health-probe kernels.
+
+Import `get_ops()` to obtain the compiled extension. On a GPU box the native
+extension is REQUIRED: failure to load raises (no silent eager fallback).
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+_ops = None
+
+
+def get_ops():
+    """Load the hived_ops HIP extension, building it if necessary."""
+    global _ops
+    if _ops is None:
+        from .build import build
+
+        _ops = build(verbose=False)
+    return _ops
+
+
+def gpu_health_report(device: int = 0, quick: bool = True) -> dict:
+    """Run the health-probe kernels on one GPU and return measured facts.
+
+    Feeds leaf-cell healthiness: HBM bandwidth deficit or MFMA mismatch marks
+    the GPU's leaf cell bad.
+    """
+    import torch
+
+    ops = get_ops()
+    torch.cuda.set_device(device)
+    size_mb = 256 if quick else 2048
+    iters = 5 if quick else 20
+    report = {"device": device, "info": ops.device_info(device)}
+    report["hbm_gbps"] = ops.hbm_triad_gbps(size_mb, iters)
+
+    torch.manual_seed(0)
+    A = (torch.randn(16, 32, dtype=torch.float32) / 8).bfloat16().cuda(device)
+    B = (torch.randn(32, 16, dtype=torch.float32) / 8).bfloat16().cuda(device)
+    tiles = ops.mfma_check(A, B, 2048, 1)  # 8192 waves >> 256 CUs
+    ref = (A.float() @ B.float())
+    max_err = (tiles - ref.unsqueeze(0)).abs().max().item()
+    tile_spread = (tiles - tiles[0].unsqueeze(0)).abs().max().item()
+    report["mfma_max_err_vs_fp32"] = max_err
+    report["mfma_cross_cu_spread"] = tile_spread  # must be exactly 0
+    report["mfma_ok"] = bool(tile_spread == 0.0 and max_err < 0.1)
+    report["healthy"] = bool(report["mfma_ok"] and report["hbm_gbps"] > 1000.0)
+    return report

@@ -1,0 +1,218 @@
+// MI355X (gfx950) GPU health-probe kernels.
+//
+// The scheduler grounds cell healthiness in measured hardware facts (the
+// reference trusts K8s NodeReady only). These kernels give
+// per-GPU signals that feed leaf-cell health:
+//  - hbm_triad:  streaming HBM3E bandwidth (expected ~6 TB/s class; a sick
+//                stack shows up as a large deficit)
+//  - mfma_check: bf16 MFMA tile GEMM on every CU, output verified bitwise by
+//                the host against a reference — catches broken matrix cores
+//  - p2p copy:   xGMI link bandwidth between two GPUs (expected ~153 GB/s per
+//                link per direction); a degraded link marks the PAIR cell bad
+//
+// Written CDNA4-native: 64-wide wavefronts, float4 coalesced loads,
+// __builtin_amdgcn_mfma_f32_16x16x32_bf16 per-wave tiles.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#define HIP_CHECK(cmd)                                                                     \
+  do {                                                                                     \
+    hipError_t e = (cmd);                                                                  \
+    TORCH_CHECK(e == hipSuccess, "HIP error: ", hipGetErrorString(e), " at ", __FILE__,    \
+                ":", __LINE__);                                                            \
+  } while (0)
+
+// ---------------------------------------------------------------------------
+// HBM streaming bandwidth: c = a + 2*b (triad), float4 per lane.
+// ---------------------------------------------------------------------------
+__global__ void triad_kernel(const float4* __restrict__ a, const float4* __restrict__ b,
+                             float4* __restrict__ c, long n) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float4 av = a[i];
+    float4 bv = b[i];
+    float4 cv;
+    cv.x = av.x + 2.0f * bv.x;
+    cv.y = av.y + 2.0f * bv.y;
+    cv.z = av.z + 2.0f * bv.z;
+    cv.w = av.w + 2.0f * bv.w;
+    c[i] = cv;
+  }
+}
+
+// Returns achieved GB/s (3 streams: 2 reads + 1 write).
+double hbm_triad_gbps(long size_mb, long iters) {
+  TORCH_CHECK(size_mb > 0 && iters > 0);
+  long bytes = size_mb * 1024 * 1024;
+  long n = bytes / sizeof(float4);
+  auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA);
+  auto a = torch::ones({bytes / 4}, opts);
+  auto b = torch::ones({bytes / 4}, opts);
+  auto c = torch::empty({bytes / 4}, opts);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  int blocks = 8192;  // >> 256 CUs to fill all 8 XCDs
+  int threads = 256;
+  // warmup
+  hipLaunchKernelGGL(triad_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     reinterpret_cast<const float4*>(a.data_ptr<float>()),
+                     reinterpret_cast<const float4*>(b.data_ptr<float>()),
+                     reinterpret_cast<float4*>(c.data_ptr<float>()), n);
+  hipEvent_t start, stop;
+  HIP_CHECK(hipEventCreate(&start));
+  HIP_CHECK(hipEventCreate(&stop));
+  HIP_CHECK(hipEventRecord(start, stream));
+  for (long it = 0; it < iters; it++) {
+    hipLaunchKernelGGL(triad_kernel, dim3(blocks), dim3(threads), 0, stream,
+                       reinterpret_cast<const float4*>(a.data_ptr<float>()),
+                       reinterpret_cast<const float4*>(b.data_ptr<float>()),
+                       reinterpret_cast<float4*>(c.data_ptr<float>()), n);
+  }
+  HIP_CHECK(hipEventRecord(stop, stream));
+  HIP_CHECK(hipEventSynchronize(stop));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, start, stop));
+  HIP_CHECK(hipEventDestroy(start));
+  HIP_CHECK(hipEventDestroy(stop));
+  double sec = ms / 1e3;
+  return (3.0 * bytes * iters) / sec / 1e9;
+}
+
+// ---------------------------------------------------------------------------
+// MFMA health check: every wave computes the same 16x16 = (16x32)x(32x16)
+// bf16 tile with v_mfma_f32_16x16x32_bf16 and writes its result; the host
+// compares all tiles against a reference. Grid >> 256 CUs so every CU's
+// matrix pipes execute it.
+//
+// Fragment layout (gfx950, 16x16x32 bf16):
+//   A: lane l holds A[m = l&15][k = (l>>4)*8 + j], j in [0,8)
+//   B: lane l holds B[k = (l>>4)*8 + j][n = l&15]
+//   C/D: lane l reg r -> C[row = (l>>4)*4 + r][col = l&15]
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) short frag8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__global__ void mfma_check_kernel(const short* __restrict__ A, const short* __restrict__ B,
+                                  float* __restrict__ C, int repeats) {
+  int lane = threadIdx.x & 63;
+  int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int m = lane & 15;
+  int kbase = (lane >> 4) * 8;
+  frag8 a, b;
+  for (int j = 0; j < 8; j++) {
+    a[j] = A[m * 32 + kbase + j];
+    b[j] = B[(kbase + j) * 16 + m];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int r = 0; r < repeats; r++) {
+    f32x4 t = {0.f, 0.f, 0.f, 0.f};
+    t = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, t, 0, 0, 0);
+    acc = t;
+  }
+  float* out = C + (size_t)wave * 256;
+  for (int r = 0; r < 4; r++) {
+    int row = (lane >> 4) * 4 + r;
+    out[row * 16 + m] = acc[r];
+  }
+}
+
+// A: [16,32] bf16, B: [32,16] bf16 -> returns [waves, 16, 16] fp32 tiles.
+torch::Tensor mfma_check(torch::Tensor A, torch::Tensor B, long blocks, long repeats) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda(), "A/B must be on GPU");
+  TORCH_CHECK(A.dtype() == torch::kBFloat16 && B.dtype() == torch::kBFloat16);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
+              B.sizes() == torch::IntArrayRef({32, 16}));
+  A = A.contiguous();
+  B = B.contiguous();
+  int threads = 256;
+  long waves = blocks * (threads / 64);
+  auto C = torch::empty({waves, 16, 16},
+                        torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(mfma_check_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     reinterpret_cast<const short*>(A.data_ptr()),
+                     reinterpret_cast<const short*>(B.data_ptr()), C.data_ptr<float>(),
+                     (int)repeats);
+  HIP_CHECK(hipGetLastError());
+  return C;
+}
+
+// ---------------------------------------------------------------------------
+// xGMI p2p bandwidth between two devices (one direction).
+// ---------------------------------------------------------------------------
+double p2p_gbps(int src_dev, int dst_dev, long size_mb, long iters) {
+  TORCH_CHECK(size_mb > 0 && iters > 0);
+  long bytes = size_mb * 1024 * 1024;
+  int canAccess = 0;
+  HIP_CHECK(hipDeviceCanAccessPeer(&canAccess, dst_dev, src_dev));
+  void *src = nullptr, *dst = nullptr;
+  HIP_CHECK(hipSetDevice(src_dev));
+  HIP_CHECK(hipMalloc(&src, bytes));
+  HIP_CHECK(hipSetDevice(dst_dev));
+  HIP_CHECK(hipMalloc(&dst, bytes));
+  if (canAccess) {
+    HIP_CHECK(hipSetDevice(dst_dev));
+    hipError_t e = hipDeviceEnablePeerAccess(src_dev, 0);
+    TORCH_CHECK(e == hipSuccess || e == hipErrorPeerAccessAlreadyEnabled,
+                "enable peer access failed: ", hipGetErrorString(e));
+  }
+  HIP_CHECK(hipSetDevice(src_dev));
+  hipStream_t stream;
+  HIP_CHECK(hipStreamCreate(&stream));
+  HIP_CHECK(hipMemcpyPeerAsync(dst, dst_dev, src, src_dev, bytes, stream));  // warmup
+  hipEvent_t start, stop;
+  HIP_CHECK(hipEventCreate(&start));
+  HIP_CHECK(hipEventCreate(&stop));
+  HIP_CHECK(hipEventRecord(start, stream));
+  for (long i = 0; i < iters; i++) {
+    HIP_CHECK(hipMemcpyPeerAsync(dst, dst_dev, src, src_dev, bytes, stream));
+  }
+  HIP_CHECK(hipEventRecord(stop, stream));
+  HIP_CHECK(hipEventSynchronize(stop));
+  float ms = 0;
+  HIP_CHECK(hipEventElapsedTime(&ms, start, stop));
+  HIP_CHECK(hipEventDestroy(start));
+  HIP_CHECK(hipEventDestroy(stop));
+  HIP_CHECK(hipStreamDestroy(stream));
+  HIP_CHECK(hipFree(src));
+  HIP_CHECK(hipSetDevice(dst_dev));
+  HIP_CHECK(hipFree(dst));
+  HIP_CHECK(hipSetDevice(src_dev));
+  double sec = ms / 1e3;
+  return (double)bytes * iters / sec / 1e9;
+}
+
+py::dict device_info(int dev) {
+  hipDeviceProp_t prop;
+  HIP_CHECK(hipGetDeviceProperties(&prop, dev));
+  py::dict d;
+  d["name"] = std::string(prop.name);
+  d["gcnArchName"] = std::string(prop.gcnArchName);
+  d["multiProcessorCount"] = prop.multiProcessorCount;
+  d["totalGlobalMem"] = (long long)prop.totalGlobalMem;
+  d["clockRate_kHz"] = prop.clockRate;
+  d["warpSize"] = prop.warpSize;
+  return d;
+}
+
+int device_count() {
+  int n = 0;
+  hipError_t e = hipGetDeviceCount(&n);
+  if (e != hipSuccess) return 0;
+  return n;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X GPU health-probe kernels (gfx950 HIP)";
+  m.def("hbm_triad_gbps", &hbm_triad_gbps, py::arg("size_mb") = 1024, py::arg("iters") = 10,
+        "Streaming HBM bandwidth in GB/s (triad: 2 reads + 1 write)");
+  m.def("mfma_check", &mfma_check, py::arg("A"), py::arg("B"), py::arg("blocks") = 2048,
+        py::arg("repeats") = 1, "Per-wave bf16 MFMA tile GEMM across all CUs");
+  m.def("p2p_gbps", &p2p_gbps, py::arg("src_dev"), py::arg("dst_dev"), py::arg("size_mb") = 256,
+        py::arg("iters") = 10, "xGMI peer-to-peer copy bandwidth in GB/s");
+  m.def("device_info", &device_info, py::arg("dev") = 0);
+  m.def("device_count", &device_count);
+}
