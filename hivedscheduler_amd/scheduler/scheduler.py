@@ -1,0 +1,338 @@
+"""HivedScheduler: the K8s bridge (L4).
+
+Parity with reference pkg/scheduler/scheduler.go: node/pod informer callbacks
+(l.132-173), PodScheduleStatuses state machine (l.110-115), filter/bind/
+preempt routines (l.485-721), force-bind escape hatch (l.423-483), recovery of
+bound pods (l.306-337). A single scheduler lock serializes scheduling.
+
+Adds a Prometheus metrics layer (per-decision latency histogram + counters),
+which the reference lacks.
+"""
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+from ..algorithm import FILTERING, PREEMPTING, HivedAlgorithm, ScheduleResult
+from ..api import constants
+from ..api.types import Config, PodBindInfo, WebServerError
+from ..internal import pod as podutil
+from ..internal.types import (
+    POD_BINDING,
+    POD_BOUND,
+    POD_PREEMPTING,
+    POD_WAITING,
+    PodScheduleStatus,
+)
+
+log = logging.getLogger("hivedscheduler")
+
+try:
+    from prometheus_client import Counter, Histogram
+
+    _LATENCY = Histogram(
+        "hived_schedule_seconds",
+        "Latency of one extender decision",
+        ["verb"],
+        buckets=(1e-5, 2e-5, 5e-5, 1e-4, 2.5e-4, 5e-4, 1e-3, 5e-3, 2.5e-2, 0.1, 1.0),
+    )
+    _DECISIONS = Counter("hived_decisions_total", "Scheduling decisions", ["kind"])
+    _HAVE_PROM = True
+except ImportError:  # pragma: no cover
+    _HAVE_PROM = False
+
+
+class HivedScheduler:
+    """Bridges informer events and the extender HTTP API to the algorithm."""
+
+    def __init__(self, config: Config, k8s_client=None):
+        self.config = config
+        self.k8s = k8s_client
+        self.algorithm = HivedAlgorithm(config)
+        self.pod_statuses: Dict[str, PodScheduleStatus] = {}  # by pod UID
+        self.lock = threading.RLock()
+        self.synced = threading.Event()
+        self.force_bind_threshold = config.forcePodBindThreshold
+        self.waiting_block_ms = config.waitingPodSchedulingBlockMilliSec
+        # nodes known to the informer (name -> node dict)
+        self.nodes: Dict[str, dict] = {}
+
+    # ------------------------------------------------------------------
+    # Informer callbacks
+    # ------------------------------------------------------------------
+    def on_node_add(self, node: dict) -> None:
+        with self.lock:
+            name = node["metadata"]["name"]
+            self.nodes[name] = node
+            self.algorithm.update_node(name, podutil.is_node_healthy(node))
+
+    def on_node_update(self, old: dict, new: dict) -> None:
+        with self.lock:
+            name = new["metadata"]["name"]
+            self.nodes[name] = new
+            old_h = podutil.is_node_healthy(old)
+            new_h = podutil.is_node_healthy(new)
+            if old_h != new_h:
+                self.algorithm.update_node(name, new_h)
+
+    def on_node_delete(self, node: dict) -> None:
+        with self.lock:
+            name = node["metadata"]["name"]
+            self.nodes.pop(name, None)
+            self.algorithm.delete_node(name)
+
+    def on_pod_add(self, pod: dict) -> None:
+        if not podutil.is_interested(pod):
+            return
+        with self.lock:
+            if podutil.is_bound(pod):
+                self._add_bound_pod(pod)
+
+    def on_pod_update(self, old: dict, new: dict) -> None:
+        old_interested = podutil.is_interested(old)
+        new_interested = podutil.is_interested(new)
+        if old_interested and not new_interested:
+            self.on_pod_delete(old)
+        elif new_interested:
+            with self.lock:
+                if podutil.is_bound(new):
+                    uid = podutil.pod_uid(new)
+                    st = self.pod_statuses.get(uid)
+                    if st is None or st.state != POD_BOUND:
+                        self._add_bound_pod(new)
+
+    def on_pod_delete(self, pod: dict) -> None:
+        if not podutil.is_interested(pod):
+            return
+        with self.lock:
+            uid = podutil.pod_uid(pod)
+            key = podutil.pod_key(pod)
+            st = self.pod_statuses.pop(uid, None)
+            if podutil.is_bound(pod) or (st is not None and st.state in (POD_BINDING, POD_BOUND)):
+                try:
+                    spec = podutil.extract_pod_scheduling_spec(pod)
+                    if st is not None and st.pod_bind_info is not None:
+                        info = st.pod_bind_info
+                    else:
+                        info = podutil.extract_pod_bind_info(pod)
+                    self.algorithm.delete_allocated_pod(spec, info, key)
+                except WebServerError as e:
+                    log.warning("[%s]: delete allocated pod failed: %s", key, e)
+            else:
+                try:
+                    spec = podutil.extract_pod_scheduling_spec(pod)
+                    self.algorithm.delete_unallocated_pod(spec, key)
+                except WebServerError as e:
+                    log.warning("[%s]: delete unallocated pod failed: %s", key, e)
+
+    def _add_bound_pod(self, pod: dict) -> None:
+        """Recovery path: rebuild allocation state from the pod-bind-info
+        annotation (the pods ARE the database)."""
+        key = podutil.pod_key(pod)
+        uid = podutil.pod_uid(pod)
+        spec = podutil.extract_pod_scheduling_spec(pod)
+        info = podutil.extract_pod_bind_info(pod)
+        self.algorithm.add_allocated_pod(spec, info, key)
+        self.pod_statuses[uid] = PodScheduleStatus(
+            pod=pod, state=POD_BOUND, pod_scheduling_spec=spec, pod_bind_info=info,
+            node=pod["spec"].get("nodeName", ""))
+
+    # ------------------------------------------------------------------
+    # Extender verbs
+    # ------------------------------------------------------------------
+    def _admission_check(self, uid: str) -> PodScheduleStatus:
+        st = self.pod_statuses.get(uid)
+        if st is None:
+            raise WebServerError.bad_request(
+                "Pod does not exist, completed or has not been informed to the scheduler")
+        if st.state == POD_BOUND:
+            raise WebServerError.bad_request(
+                f"Pod has already been bound to node {st.node}")
+        return st
+
+    def _should_force_bind(self, st: PodScheduleStatus, suggested: List[str]) -> bool:
+        if st.pod_bind_attempts >= self.force_bind_threshold:
+            return True
+        node = st.node
+        # decision already invalid w.r.t. current state: bind now, let pod-side
+        # failure drive the retry
+        if self.nodes and node not in self.nodes:
+            return True
+        if suggested and node not in suggested:
+            return True
+        return False
+
+    def _force_bind(self, st: PodScheduleStatus) -> None:
+        def _run():
+            try:
+                self.bind({
+                    "PodName": st.pod["metadata"]["name"],
+                    "PodNamespace": st.pod["metadata"].get("namespace", "default"),
+                    "PodUID": podutil.pod_uid(st.pod),
+                    "Node": st.node,
+                })
+            except Exception as e:  # force bind is best-effort
+                log.warning("force bind failed: %s", e)
+
+        threading.Thread(target=_run, name="force-bind", daemon=True).start()
+
+    def filter(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        t0 = time.perf_counter()
+        try:
+            return self._filter(args)
+        finally:
+            if _HAVE_PROM:
+                _LATENCY.labels("filter").observe(time.perf_counter() - t0)
+
+    def _filter(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        with self.lock:
+            pod = args["Pod"]
+            suggested = list(args.get("NodeNames") or [])
+            uid = podutil.pod_uid(pod)
+            key = podutil.pod_key(pod)
+
+            # pods are tracked lazily: first filter call registers them
+            if uid not in self.pod_statuses:
+                self.pod_statuses[uid] = PodScheduleStatus(pod=pod, state=POD_WAITING)
+            st = self._admission_check(uid)
+            if st.state == POD_BINDING:
+                # insist the previous decision; binding is idempotent
+                st.pod_bind_attempts += 1
+                if self._should_force_bind(st, suggested):
+                    self._force_bind(st)
+                return {"NodeNames": [st.node]}
+
+            spec = podutil.extract_pod_scheduling_spec(pod)
+            result = self.algorithm.schedule(spec, key, suggested, FILTERING)
+            if result.kind == "bind":
+                info = result.bind_info
+                binding_pod = podutil.new_binding_pod(pod, info)
+                # optimistic commit: assume allocated before the real bind
+                self.algorithm.add_allocated_pod(spec, info, key)
+                st = PodScheduleStatus(
+                    pod=binding_pod, state=POD_BINDING, pod_scheduling_spec=spec,
+                    pod_bind_info=info, node=info.node)
+                self.pod_statuses[uid] = st
+                if _HAVE_PROM:
+                    _DECISIONS.labels("bind").inc()
+                if self._should_force_bind(st, suggested):
+                    self._force_bind(st)
+                return {"NodeNames": [info.node]}
+            if result.kind == "preempt":
+                if _HAVE_PROM:
+                    _DECISIONS.labels("preempt").inc()
+                failed = {
+                    result.victim_node: "node(%s) has preemptible Pods: %s" % (
+                        result.victim_node, ", ".join(result.victim_pod_keys))
+                }
+                return {"FailedNodes": failed}
+            # wait
+            if _HAVE_PROM:
+                _DECISIONS.labels("wait").inc()
+            self.pod_statuses[uid] = PodScheduleStatus(
+                pod=pod, state=POD_WAITING, pod_scheduling_spec=spec)
+            if self.waiting_block_ms > 0:
+                time.sleep(self.waiting_block_ms / 1e3)
+            reason = "Pod is waiting for preemptible or free resource to appear"
+            if result.wait_reason:
+                reason += ": " + result.wait_reason
+            return {"FailedNodes": {constants.ComponentName: reason}}
+
+    def bind(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        t0 = time.perf_counter()
+        try:
+            return self._bind(args)
+        finally:
+            if _HAVE_PROM:
+                _LATENCY.labels("bind").observe(time.perf_counter() - t0)
+
+    def _bind(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        with self.lock:
+            uid = str(args.get("PodUID", ""))
+            node = args.get("Node", "")
+            st = self._admission_check(uid)
+            if st.state != POD_BINDING:
+                raise WebServerError.bad_request(
+                    f"Pod cannot be bound without a scheduling placement: state {st.state}, "
+                    f"received node {node}")
+            if st.node != node:
+                raise WebServerError.bad_request(
+                    f"Pod binding node mismatch: expected {st.node}, received {node}")
+            if self.k8s is not None:
+                # the K8s Bind subresource is atomic and at-most-once
+                ann = st.pod["metadata"].get("annotations", {})
+                self.k8s.bind_pod(
+                    namespace=st.pod["metadata"].get("namespace", "default"),
+                    name=st.pod["metadata"]["name"],
+                    uid=uid,
+                    node=node,
+                    annotations={
+                        constants.AnnotationKeyPodLeafCellIsolation:
+                            ann.get(constants.AnnotationKeyPodLeafCellIsolation, ""),
+                        constants.AnnotationKeyPodBindInfo:
+                            ann.get(constants.AnnotationKeyPodBindInfo, ""),
+                    },
+                )
+            st.transition(POD_BOUND)
+            return {}
+
+    def preempt(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        t0 = time.perf_counter()
+        try:
+            return self._preempt(args)
+        finally:
+            if _HAVE_PROM:
+                _LATENCY.labels("preempt").observe(time.perf_counter() - t0)
+
+    def _preempt(self, args: Dict[str, Any]) -> Dict[str, Any]:
+        with self.lock:
+            pod = args["Pod"]
+            uid = podutil.pod_uid(pod)
+            key = podutil.pod_key(pod)
+            suggested = list((args.get("NodeNameToMetaVictims") or
+                              args.get("NodeNameToVictims") or {}).keys())
+            if uid not in self.pod_statuses:
+                self.pod_statuses[uid] = PodScheduleStatus(pod=pod, state=POD_WAITING)
+            st = self._admission_check(uid)
+            if st.state == POD_BINDING:
+                raise WebServerError.bad_request(
+                    f"Pod has already been binding to node {st.node}")
+
+            spec = podutil.extract_pod_scheduling_spec(pod)
+            result = self.algorithm.schedule(spec, key, suggested, PREEMPTING)
+            if result.kind == "bind":
+                # do not bind here; let the next filter call do it
+                return {}
+            if result.kind == "preempt":
+                self.pod_statuses[uid] = PodScheduleStatus(
+                    pod=pod, state=POD_PREEMPTING, pod_scheduling_spec=spec,
+                    victim_pod_keys=result.victim_pod_keys)
+                victims = {result.victim_node: {
+                    "Pods": [{"UID": k} for k in result.victim_pod_keys]}}
+                return {"NodeNameToMetaVictims": victims}
+            self.pod_statuses[uid] = PodScheduleStatus(
+                pod=pod, state=POD_WAITING, pod_scheduling_spec=spec)
+            return {}
+
+    # ------------------------------------------------------------------
+    # Inspect delegation
+    # ------------------------------------------------------------------
+    def get_all_affinity_groups(self):
+        return self.algorithm.get_all_affinity_groups()
+
+    def get_affinity_group(self, name: str):
+        return self.algorithm.get_affinity_group(name)
+
+    def get_cluster_status(self):
+        return self.algorithm.get_cluster_status()
+
+    def get_physical_cluster_status(self):
+        return self.algorithm.get_physical_cluster_status()
+
+    def get_all_virtual_clusters_status(self):
+        return self.algorithm.get_all_virtual_clusters_status()
+
+    def get_virtual_cluster_status(self, vc: str):
+        return self.algorithm.get_virtual_cluster_status(vc)

@@ -1,0 +1,117 @@
+"""Scheduler-extender HTTP server (L5).
+
+Parity with reference pkg/webserver/webserver.go: routes (l.77-84)
+/, /v1/extender/{filter,bind,preempt}, /v1/inspect/{affinitygroups,
+clusterstatus[,physicalcluster,virtualclusters]}, JSON decode/validate,
+error->HTTP-status translation (l.142-155). Adds /metrics (Prometheus) and
+/healthz, which the reference lacks.
+
+The K8s default scheduler POSTs the extender verbs per the policy config
+(filterVerb/bindVerb/preemptVerb; examples/deploy/).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Any, Dict, Optional
+
+from fastapi import FastAPI, Request, Response
+from fastapi.responses import JSONResponse, PlainTextResponse
+
+from ..api import constants
+from ..api.types import WebServerError
+from ..scheduler import HivedScheduler
+
+log = logging.getLogger("hivedscheduler.webserver")
+
+
+def create_app(scheduler: HivedScheduler) -> FastAPI:
+    app = FastAPI(title="hivedscheduler-amd", docs_url=None, redoc_url=None)
+
+    @app.exception_handler(WebServerError)
+    async def _hived_error(request: Request, exc: WebServerError):
+        return JSONResponse(status_code=exc.code, content={"error": exc.message})
+
+    @app.get(constants.RootPath)
+    async def root():
+        return {"component": constants.ComponentName, "paths": [
+            constants.FilterPath, constants.BindPath, constants.PreemptPath,
+            constants.AffinityGroupsPath, constants.ClusterStatusPath,
+            constants.PhysicalClusterPath, constants.VirtualClustersPath,
+            constants.MetricsPath,
+        ]}
+
+    @app.get("/healthz")
+    async def healthz():
+        return PlainTextResponse("ok")
+
+    # ---- extender verbs ----
+    @app.post(constants.FilterPath)
+    async def filter_verb(args: Dict[str, Any]):
+        if not isinstance(args.get("Pod"), dict):
+            raise WebServerError.bad_request("ExtenderArgs.Pod is missing")
+        return scheduler.filter(args)
+
+    @app.post(constants.BindPath)
+    async def bind_verb(args: Dict[str, Any]):
+        for field in ("PodName", "PodNamespace", "PodUID", "Node"):
+            if not args.get(field):
+                raise WebServerError.bad_request(f"ExtenderBindingArgs.{field} is missing")
+        try:
+            return scheduler.bind(args)
+        except WebServerError as e:
+            # binding errors are returned in-band so the default scheduler
+            # surfaces them on the pod (reference webserver.go:194-215)
+            return {"Error": e.message}
+
+    @app.post(constants.PreemptPath)
+    async def preempt_verb(args: Dict[str, Any]):
+        if not isinstance(args.get("Pod"), dict):
+            raise WebServerError.bad_request("ExtenderPreemptionArgs.Pod is missing")
+        return scheduler.preempt(args)
+
+    # ---- inspect API ----
+    @app.get(constants.AffinityGroupsPath)
+    async def affinity_groups():
+        return scheduler.get_all_affinity_groups()
+
+    @app.get(constants.AffinityGroupsPath + "{name:path}")
+    async def affinity_group(name: str):
+        return scheduler.get_affinity_group(name)
+
+    @app.get(constants.ClusterStatusPath)
+    async def cluster_status():
+        return scheduler.get_cluster_status()
+
+    @app.get(constants.PhysicalClusterPath)
+    async def physical_cluster():
+        return scheduler.get_physical_cluster_status()
+
+    @app.get(constants.VirtualClustersPath)
+    async def virtual_clusters():
+        return scheduler.get_all_virtual_clusters_status()
+
+    @app.get(constants.VirtualClustersPath + "{vc}")
+    async def virtual_cluster(vc: str):
+        return scheduler.get_virtual_cluster_status(vc)
+
+    # ---- metrics ----
+    @app.get(constants.MetricsPath)
+    async def metrics():
+        try:
+            from prometheus_client import CONTENT_TYPE_LATEST, generate_latest
+
+            return Response(content=generate_latest(), media_type=CONTENT_TYPE_LATEST)
+        except ImportError:  # pragma: no cover
+            return PlainTextResponse("prometheus_client not available", status_code=501)
+
+    return app
+
+
+def run_server(scheduler: HivedScheduler, address: Optional[str] = None) -> None:
+    """Blocking uvicorn server on config.webServerAddress (":9096" style)."""
+    import uvicorn
+
+    addr = address or scheduler.config.webServerAddress or constants.DefaultWebServerAddress
+    host, _, port = addr.rpartition(":")
+    host = host or "0.0.0.0"
+    uvicorn.run(create_app(scheduler), host=host, port=int(port), log_level="info")

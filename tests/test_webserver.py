@@ -1,0 +1,88 @@
+"""Webserver (L5): extender HTTP protocol + inspect API + metrics, via the
+Starlette test client (no real server)."""
+import pytest
+import yaml
+
+from hivedscheduler_amd.api import constants
+from hivedscheduler_amd.scheduler import HivedScheduler
+from hivedscheduler_amd.sim import mi355x_cluster_config
+from hivedscheduler_amd.webserver import create_app
+
+from test_scheduler_framework import make_node, make_pod
+
+
+@pytest.fixture()
+def client():
+    from starlette.testclient import TestClient
+
+    cfg = mi355x_cluster_config(num_nodes=1)
+    sched = HivedScheduler(cfg)
+    sched.on_node_add(make_node("node1"))
+    return TestClient(create_app(sched), raise_server_exceptions=False)
+
+
+def test_root_lists_paths(client):
+    r = client.get("/")
+    assert r.status_code == 200
+    assert constants.FilterPath in r.json()["paths"]
+
+
+def test_filter_bind_over_http(client):
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    r = client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    assert r.status_code == 200
+    assert r.json()["NodeNames"] == ["node1"]
+    r = client.post(constants.BindPath, json={
+        "PodName": "p1", "PodNamespace": "ns", "PodUID": pod["metadata"]["uid"],
+        "Node": "node1"})
+    assert r.status_code == 200
+    assert not r.json().get("Error")
+
+
+def test_bind_error_in_band(client):
+    r = client.post(constants.BindPath, json={
+        "PodName": "nope", "PodNamespace": "ns", "PodUID": "u-nope", "Node": "node1"})
+    assert r.status_code == 200
+    assert r.json()["Error"]
+
+
+def test_bad_request_translation(client):
+    pod = make_pod("p1", {"virtualCluster": "NOPE", "priority": 0, "leafCellNumber": 2})
+    r = client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    assert r.status_code == 400
+    assert "NOPE" in r.json()["error"]
+    r = client.post(constants.FilterPath, json={"NodeNames": ["node1"]})
+    assert r.status_code == 400
+
+
+def test_inspect_api(client):
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    r = client.get(constants.AffinityGroupsPath)
+    assert r.status_code == 200
+    assert [g["name"] for g in r.json()] == ["ns/p1"]
+    r = client.get(constants.AffinityGroupsPath + "ns/p1")
+    assert r.status_code == 200
+    assert r.json()["vc"] == "VC1"
+    assert client.get(constants.AffinityGroupsPath + "missing").status_code == 400
+    r = client.get(constants.ClusterStatusPath)
+    assert r.status_code == 200
+    assert "physicalCluster" in r.json() and "virtualClusters" in r.json()
+    r = client.get(constants.PhysicalClusterPath)
+    assert r.status_code == 200
+    assert r.json()[0]["cellType"] == "MI355X-NODE"
+    r = client.get(constants.VirtualClustersPath + "VC1")
+    assert r.status_code == 200
+    assert client.get(constants.VirtualClustersPath + "NOPE").status_code == 404
+
+
+def test_metrics_endpoint(client):
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    r = client.get(constants.MetricsPath)
+    assert r.status_code == 200
+    assert "hived_schedule_seconds" in r.text
+
+
+def test_healthz(client):
+    assert client.get("/healthz").text == "ok"
