@@ -112,6 +112,11 @@ class HivedAlgorithm:
     def set_healthy_node(self, name: str) -> None:
         self._call(self._core.set_node_healthy, name, True)
 
+    def set_leaf_cell_healthy(self, node: str, leaf_index: int, healthy: bool) -> None:
+        """GPU/xGMI-level health: marks one leaf cell; badness rolls up to the
+        pair/quad/node cells. Independent of node-level health."""
+        self._call(self._core.set_leaf_cell_healthy, node, leaf_index, healthy)
+
     def all_nodes(self) -> List[str]:
         return self._call(self._core.all_nodes)
 

@@ -397,13 +397,36 @@ void HivedCore::setNodeHealthy(const std::string& node, bool healthy) {
     badNodes_.erase(node);
     auto it = nodeLeafCellsStorage_.find(node);
     if (it == nodeLeafCellsStorage_.end()) return;
-    for (PhysicalCell* leaf : it->second) setHealthyCell(leaf);
+    for (PhysicalCell* leaf : it->second) {
+      if (!badLeafMarks_.count(leaf)) setHealthyCell(leaf);
+    }
   } else {
     if (badNodes_.count(node)) return;
     badNodes_.insert(node);
     auto it = nodeLeafCellsStorage_.find(node);
     if (it == nodeLeafCellsStorage_.end()) return;
     for (PhysicalCell* leaf : it->second) setBadCell(leaf);
+  }
+}
+
+// Fine-grained health: one GPU (leaf cell) on a node. A degraded xGMI link is
+// reported against its endpoint leaf; the pair/quad cell turns bad via the
+// any-child-bad roll-up. Node-level badness (badNodes_) takes precedence: a
+// leaf on a bad node cannot be marked healthy.
+void HivedCore::setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy) {
+  auto it = nodeLeafCellsStorage_.find(node);
+  if (it == nodeLeafCellsStorage_.end()) return;
+  for (PhysicalCell* leaf : it->second) {
+    if (!leaf->leafIndices.empty() && leaf->leafIndices[0] == leafIndex) {
+      if (healthy) {
+        badLeafMarks_.erase(leaf);
+        if (!badNodes_.count(node)) setHealthyCell(leaf);
+      } else {
+        badLeafMarks_.insert(leaf);
+        setBadCell(leaf);
+      }
+      return;
+    }
   }
 }
 

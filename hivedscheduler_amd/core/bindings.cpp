@@ -284,6 +284,9 @@ class PyHivedCore {
   explicit PyHivedCore(const py::dict& spec) : core_(parseClusterSpec(spec)) {}
 
   void setNodeHealthy(const std::string& node, bool healthy) { core_.setNodeHealthy(node, healthy); }
+  void setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy) {
+    core_.setLeafCellHealthy(node, leafIndex, healthy);
+  }
   std::vector<std::string> allNodes() const { return core_.allNodes(); }
   std::vector<std::string> badNodes() const {
     auto s = core_.badNodes();
@@ -438,6 +441,8 @@ PYBIND11_MODULE(hivedcore, m) {
   py::class_<PyHivedCore>(m, "HivedCore")
       .def(py::init<const py::dict&>(), py::arg("spec"))
       .def("set_node_healthy", &PyHivedCore::setNodeHealthy, py::arg("node"), py::arg("healthy"))
+      .def("set_leaf_cell_healthy", &PyHivedCore::setLeafCellHealthy, py::arg("node"),
+           py::arg("leaf_index"), py::arg("healthy"))
       .def("all_nodes", &PyHivedCore::allNodes)
       .def("bad_nodes", &PyHivedCore::badNodes)
       .def("schedule", &PyHivedCore::schedule, py::arg("pod_spec"), py::arg("pod_key"),

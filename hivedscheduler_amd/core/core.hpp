@@ -315,6 +315,9 @@ class HivedCore {
 
   // -- node health (informer events) --
   void setNodeHealthy(const std::string& node, bool healthy);
+  // -- GPU/xGMI-level health (rocm-smi exporter / probe events): marks one
+  // leaf cell; badness rolls up to pair/quad/node cells automatically --
+  void setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy);
   std::vector<std::string> allNodes() const;
   std::set<std::string> badNodes() const { return badNodes_; }
 
@@ -440,6 +443,8 @@ class HivedCore {
   std::map<std::string, std::map<int, int>> allVCDoomedBadCellNum_;
 
   std::set<std::string> badNodes_;
+  // leaves individually marked bad (GPU/xGMI level), independent of node health
+  std::set<PhysicalCell*> badLeafMarks_;
   std::map<std::string, std::vector<std::string>> cellChains_;        // leaf type -> chains
   std::map<std::string, std::map<int, std::string>> cellTypes_;      // chain -> level -> type
   std::map<std::string, std::map<int, int>> leafCellNums_;           // chain -> level -> leaf num

@@ -39,11 +39,21 @@ struct PickSession {
   std::unordered_set<Cell*> taken;
 };
 
-// (availableTotal, availableFree): leaves that are free or lower-priority
-// (preemptible), excluding ones already taken in this session.
+// Leaf-granular health: a bad physical leaf (sick GPU / degraded xGMI link
+// endpoint) is never available; a virtual leaf bound to a bad physical leaf
+// (doomed-bad binding) is equally unavailable. This is finer than the
+// reference, whose health model stops at node granularity.
+bool leafHealthy(Cell* c) {
+  if (c->isPhysical()) return c->healthy;
+  PhysicalCell* p = static_cast<VirtualCell*>(c)->phys;
+  return p == nullptr || p->healthy;
+}
+
+// (availableTotal, availableFree): healthy leaves that are free or
+// lower-priority (preemptible), excluding ones already taken in this session.
 std::pair<int, int> availLeaves(Cell* c, int p, PickSession& s) {
   if (c->level == kLowestLevel) {
-    if (s.taken.count(c)) return {0, 0};
+    if (s.taken.count(c) || !leafHealthy(c)) return {0, 0};
     if (c->priority == kFreePriority) return {1, 1};
     if (c->priority < p) return {1, 0};
     return {0, 0};
@@ -133,20 +143,22 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   // (desc, packing) > higher-priority used (asc, stay away).
   std::vector<NodeView> cv;
   cv.reserve(viewCells_.size());
+  PickSession probe;  // empty: availability before any placement
   for (Cell* c : viewCells_) {
     NodeView n;
     n.c = c;
     n.usedSamePriority = c->usedAt(priority);
     n.usedHigherPriority = 0;
-    n.freeAtPriority = c->totalLeaf;
     for (auto& [p, num] : c->usedLeafAtPriority) {
       if (crossPriorityPack_) {
         if (p != priority) n.usedSamePriority += num;
       } else if (p > priority) {
         n.usedHigherPriority += num;
       }
-      if (p >= priority) n.freeAtPriority -= num;
     }
+    // health-aware availability: bad leaves never count, so partially-bad
+    // nodes stay usable for their healthy pairs/quads
+    n.freeAtPriority = availLeaves(c, priority, probe).first;
     auto [healthy, suggested] = healthyAndSuggested(c, suggestedNodes, ignoreSuggestedNodes);
     n.healthy = healthy;
     n.suggested = suggested;
@@ -166,10 +178,8 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   for (size_t nodeIndex = 0; nodeIndex < cv.size() && podIndex < sortedLeafNums.size();) {
     const NodeView& n = cv[nodeIndex];
     if (n.freeAtPriority - pickedLeafCellNum >= sortedLeafNums[podIndex]) {
-      if (!n.healthy) {
-        *failedReason = "have to use at least one bad node " + n.c->address;
-        return false;
-      }
+      // note: a partially-bad node is usable (its avail excludes bad leaves);
+      // a fully-bad node has avail 0 and is never reached
       if (!n.suggested) {
         *failedReason = "have to use at least one non-suggested node " + n.c->address;
         return false;

@@ -111,3 +111,32 @@ def test_xgmi_link_degradation_marks_pair_bad():
     sim.alg.set_healthy_node("node1")
     r = sim.schedule("ns/p", sim.pod_spec(leaf_cells=2))
     assert r.kind == "bind"
+
+
+def test_leaf_cell_level_health():
+    """A single bad GPU (or degraded xGMI link endpoint) marks its leaf bad;
+    the pair cell goes bad via roll-up; the rest of the node stays usable."""
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    sim.alg.set_leaf_cell_healthy("node1", 0, False)
+    # a pair request avoids the damaged pair
+    r = sim.schedule("ns/pair", sim.pod_spec(leaf_cells=2))
+    assert r.kind == "bind"
+    assert 0 not in r.bind_info.leafCellIsolation
+    # a full-node request cannot be served while one GPU is bad
+    assert sim.schedule("ns/full", sim.pod_spec(leaf_cells=8)).kind == "wait"
+    # recovery restores the leaf
+    sim.delete_pod("ns/pair")
+    sim.alg.set_leaf_cell_healthy("node1", 0, True)
+    assert sim.schedule("ns/full2", sim.pod_spec(leaf_cells=8)).kind == "bind"
+
+
+def test_leaf_badness_survives_node_health_cycle():
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    sim.alg.set_leaf_cell_healthy("node1", 3, False)
+    sim.alg.set_bad_node("node1")
+    sim.alg.set_healthy_node("node1")
+    # the individually-marked leaf stays bad after the node recovers
+    r = sim.schedule("ns/q", sim.pod_spec(leaf_cells=4))
+    assert r.kind == "bind"
+    assert sorted(r.bind_info.leafCellIsolation) == [4, 5, 6, 7]
+    assert sim.schedule("ns/full", sim.pod_spec(leaf_cells=8)).kind == "wait"
