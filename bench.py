@@ -108,7 +108,7 @@ def main():
 
         backend = "nccl" if has_cuda else "gloo"
         if has_cuda:
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
 
     requests = build_requests(seed=0, count=args.requests)

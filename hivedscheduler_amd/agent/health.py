@@ -1,0 +1,138 @@
+"""Node health agent (DaemonSet-style): grounds cell healthiness in measured
+CDNA4 facts and reports GPU-level health to the scheduler.
+
+Per SURVEY.md §2.2, this replaces the reference's K8s-NodeReady-only health
+signal with: rocm-smi device state, HIP health-probe kernels (HBM bandwidth,
+MFMA verification), and the RCCL xGMI pair probe. Degraded xGMI links mark
+the PAIR cell bad (via its endpoint leaf), not the whole node.
+
+Runs on each GPU node: `python -m hivedscheduler_amd.agent --scheduler URL`.
+Reports POST to the scheduler's /v1/health/nodes/{node} endpoint:
+  {"gpus": {"0": {"healthy": true, ...}, ...}}
+"""
+from __future__ import annotations
+
+import json
+import logging
+import os
+import socket
+import subprocess
+import time
+from typing import Dict, List, Optional
+
+log = logging.getLogger("hivedscheduler.agent")
+
+
+def _rocm_smi_gpu_state() -> Dict[int, bool]:
+    """GPU liveness via rocm-smi JSON (a GPU missing or erroring is bad)."""
+    try:
+        out = subprocess.run(["rocm-smi", "--showuse", "--json"], capture_output=True,
+                             text=True, timeout=60)
+        if out.returncode != 0:
+            return {}
+        data = json.loads(out.stdout)
+        state = {}
+        for key in data:
+            if key.lower().startswith("card"):
+                idx = int(key[4:])
+                state[idx] = True
+        return state
+    except Exception as e:
+        log.warning("rocm-smi query failed: %s", e)
+        return {}
+
+
+def collect_node_health(
+    deep: bool = False,
+    min_hbm_gbps: float = 2000.0,
+    min_pair_busbw_gbps: float = 50.0,
+    probe_pairs: bool = False,
+) -> dict:
+    """One health sweep over all visible GPUs. deep=True also runs the HIP
+    kernels (HBM + MFMA); probe_pairs=True RCCL-probes each xGMI pair."""
+    report: dict = {"node": socket.gethostname(), "time": time.time(), "gpus": {}}
+    alive = _rocm_smi_gpu_state()
+    try:
+        import torch
+
+        n = torch.cuda.device_count() if torch.cuda.is_available() else 0
+    except Exception:
+        n = 0
+    n = max(n, len(alive))
+    for i in range(n):
+        gpu = {"healthy": alive.get(i, True)}
+        report["gpus"][str(i)] = gpu
+    if deep and n > 0:
+        from ..ops import gpu_health_report
+
+        for i in range(n):
+            try:
+                rep = gpu_health_report(i, quick=True)
+                report["gpus"][str(i)].update(
+                    hbm_gbps=round(rep["hbm_gbps"], 1),
+                    mfma_ok=rep["mfma_ok"],
+                )
+                if not rep["mfma_ok"] or rep["hbm_gbps"] < min_hbm_gbps:
+                    report["gpus"][str(i)]["healthy"] = False
+            except Exception as e:
+                report["gpus"][str(i)].update(healthy=False, error=str(e)[:200])
+    if probe_pairs and n >= 2:
+        from ..probe import CellProbeRunner
+
+        runner = CellProbeRunner(min_busbw_gbps=min_pair_busbw_gbps)
+        if runner.available():
+            report["pairs"] = {}
+            for p in range(n // 2):
+                pair = [2 * p, 2 * p + 1]
+                try:
+                    res = runner.probe_cell(pair, size_mb=64, iters=10)
+                    report["pairs"][f"{pair[0]}-{pair[1]}"] = res
+                    if res.get("ok") and not res.get("healthy", True):
+                        # degraded xGMI link: mark both endpoints' leaves bad
+                        for g in pair:
+                            report["gpus"][str(g)]["healthy"] = False
+                            report["gpus"][str(g)]["xgmi_degraded"] = True
+                except Exception as e:
+                    report["pairs"][f"{pair[0]}-{pair[1]}"] = {"ok": False, "error": str(e)[:200]}
+    return report
+
+
+class NodeHealthAgent:
+    """Periodic health loop posting reports to the scheduler."""
+
+    def __init__(self, scheduler_url: str, node_name: Optional[str] = None,
+                 interval_s: float = 60.0, deep_every: int = 10,
+                 probe_pairs: bool = False):
+        self.scheduler_url = scheduler_url.rstrip("/")
+        self.node_name = node_name or socket.gethostname()
+        self.interval_s = interval_s
+        self.deep_every = deep_every
+        self.probe_pairs = probe_pairs
+        self._rounds = 0
+
+    def post_report(self, report: dict) -> bool:
+        import requests
+
+        url = f"{self.scheduler_url}/v1/health/nodes/{self.node_name}"
+        try:
+            r = requests.post(url, json=report, timeout=30)
+            return r.status_code < 300
+        except Exception as e:
+            log.warning("health report post failed: %s", e)
+            return False
+
+    def run_once(self) -> dict:
+        deep = (self._rounds % self.deep_every) == 0
+        self._rounds += 1
+        report = collect_node_health(deep=deep, probe_pairs=self.probe_pairs and deep)
+        report["node"] = self.node_name
+        self.post_report(report)
+        return report
+
+    def run_forever(self) -> None:  # pragma: no cover
+        while True:
+            try:
+                self.run_once()
+            except Exception as e:
+                log.error("health sweep failed: %s", e)
+            time.sleep(self.interval_s)

@@ -58,6 +58,8 @@ class HivedScheduler:
         self.waiting_block_ms = config.waitingPodSchedulingBlockMilliSec
         # nodes known to the informer (name -> node dict)
         self.nodes: Dict[str, dict] = {}
+        # latest GPU-level health report per node (from the node agents)
+        self.health_reports: Dict[str, dict] = {}
 
     # ------------------------------------------------------------------
     # Informer callbacks
@@ -126,6 +128,26 @@ class HivedScheduler:
                     self.algorithm.delete_unallocated_pod(spec, key)
                 except WebServerError as e:
                     log.warning("[%s]: delete unallocated pod failed: %s", key, e)
+
+    def on_health_report(self, node: str, report: dict) -> dict:
+        """GPU-level health from a node agent: marks individual leaf cells
+        bad/healthy (xGMI-degraded pairs arrive as bad endpoint leaves)."""
+        with self.lock:
+            self.health_reports[node] = report
+            applied = {}
+            for idx_str, gpu in (report.get("gpus") or {}).items():
+                try:
+                    idx = int(idx_str)
+                except (TypeError, ValueError):
+                    continue
+                healthy = bool(gpu.get("healthy", True))
+                self.algorithm.set_leaf_cell_healthy(node, idx, healthy)
+                applied[idx_str] = healthy
+            return {"node": node, "applied": applied}
+
+    def get_health_reports(self) -> Dict[str, dict]:
+        with self.lock:
+            return dict(self.health_reports)
 
     def _add_bound_pod(self, pod: dict) -> None:
         """Recovery path: rebuild allocation state from the pod-bind-info

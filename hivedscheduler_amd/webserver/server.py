@@ -94,6 +94,15 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
     async def virtual_cluster(vc: str):
         return scheduler.get_virtual_cluster_status(vc)
 
+    # ---- GPU-level health intake (node agents) ----
+    @app.post("/v1/health/nodes/{node}")
+    async def health_report(node: str, report: Dict[str, Any]):
+        return scheduler.on_health_report(node, report)
+
+    @app.get("/v1/inspect/health")
+    async def health_reports():
+        return scheduler.get_health_reports()
+
     # ---- metrics ----
     @app.get(constants.MetricsPath)
     async def metrics():

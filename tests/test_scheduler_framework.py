@@ -198,3 +198,14 @@ def test_force_bind_on_threshold(sched):
         time.sleep(0.05)
     assert sched.pod_statuses[uid].state == "Bound"
     assert sched.pod_statuses[uid].node == node
+
+
+def test_health_report_marks_leaf_cells(sched):
+    sched.on_health_report("node1", {"gpus": {str(i): {"healthy": False} for i in range(8)}})
+    pod = make_pod("p8", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 8})
+    r = sched.filter({"Pod": pod, "NodeNames": ["node1", "node2"]})
+    assert r["NodeNames"] == ["node2"]
+    # recovery
+    sched.on_health_report("node1", {"gpus": {str(i): {"healthy": True} for i in range(8)}})
+    pod2 = make_pod("p8b", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 8})
+    assert sched.filter({"Pod": pod2, "NodeNames": ["node1", "node2"]})["NodeNames"] == ["node1"]

@@ -86,3 +86,18 @@ def test_metrics_endpoint(client):
 
 def test_healthz(client):
     assert client.get("/healthz").text == "ok"
+
+
+def test_health_report_endpoint(client):
+    # one bad GPU reported by the node agent
+    r = client.post("/v1/health/nodes/node1", json={
+        "gpus": {"0": {"healthy": False, "hbm_gbps": 900.0},
+                 "1": {"healthy": True}}})
+    assert r.status_code == 200
+    assert r.json()["applied"] == {"0": False, "1": True}
+    # placement now avoids GPU 0
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    rr = client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    assert rr.status_code == 200 and rr.json().get("NodeNames")
+    reports = client.get("/v1/inspect/health").json()
+    assert "node1" in reports
