@@ -1,0 +1,30 @@
+"""Example configs parse and construct a working scheduler."""
+import os
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_example_cluster_config():
+    from hivedscheduler_amd.api import config as apicfg
+    from hivedscheduler_amd.sim import SimScheduler
+
+    cfg = apicfg.init_raw_config(os.path.join(REPO, "examples/config/mi355x-cluster.yaml"))
+    sim = SimScheduler(cfg)
+    r = sim.schedule("ns/j1", sim.pod_spec(vc="prod", leaf_cells=8))
+    assert r.kind == "bind"
+    r2 = sim.schedule("ns/j2", sim.pod_spec(vc="research", leaf_cells=4))
+    assert r2.kind == "bind"
+
+
+def test_standalone_entry_config_parses():
+    # the deploy ConfigMap's embedded config also parses
+    import yaml
+    from hivedscheduler_amd.api.config import new_config
+    from hivedscheduler_amd.algorithm import HivedAlgorithm
+
+    docs = list(yaml.safe_load_all(open(os.path.join(REPO, "examples/deploy/hivedscheduler.yaml"))))
+    cm = [d for d in docs if d and d.get("kind") == "ConfigMap"
+          and d["metadata"]["name"] == "hivedscheduler-config"][0]
+    cfg = new_config(yaml.safe_load(cm["data"]["hivedscheduler.yaml"]))
+    alg = HivedAlgorithm(cfg)
+    assert alg.all_nodes() == ["mi355x-node-1"]
