@@ -121,12 +121,43 @@ class NodeHealthAgent:
             log.warning("health report post failed: %s", e)
             return False
 
+    def run_placement_probes(self) -> List[dict]:
+        """Poll post-bind probe tasks for this node, run the native RCCL
+        probe over each placement's GPUs, and post the results back."""
+        import requests
+
+        from ..probe import CellProbeRunner
+
+        url = f"{self.scheduler_url}/v1/health/probes/{self.node_name}"
+        try:
+            tasks = requests.get(url, timeout=30).json()
+        except Exception as e:
+            log.warning("probe poll failed: %s", e)
+            return []
+        runner = CellProbeRunner()
+        results = []
+        for task in tasks:
+            cells = [int(i) for i in task.get("leafCellIndices", [])]
+            if runner.available():
+                res = runner.probe_cell(cells, size_mb=64, iters=10)
+            else:
+                res = {"ok": False, "error": "rccl-cell-probe binary missing"}
+            res.update(group=task.get("group", ""), node=self.node_name,
+                       leafCellIndices=cells)
+            try:
+                requests.post(f"{self.scheduler_url}/v1/health/probes", json=res, timeout=30)
+            except Exception as e:
+                log.warning("probe result post failed: %s", e)
+            results.append(res)
+        return results
+
     def run_once(self) -> dict:
         deep = (self._rounds % self.deep_every) == 0
         self._rounds += 1
         report = collect_node_health(deep=deep, probe_pairs=self.probe_pairs and deep)
         report["node"] = self.node_name
         self.post_report(report)
+        report["placement_probes"] = self.run_placement_probes()
         return report
 
     def run_forever(self) -> None:  # pragma: no cover

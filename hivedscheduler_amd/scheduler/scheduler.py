@@ -60,6 +60,10 @@ class HivedScheduler:
         self.nodes: Dict[str, dict] = {}
         # latest GPU-level health report per node (from the node agents)
         self.health_reports: Dict[str, dict] = {}
+        # post-bind RCCL placement validation
+        from .probe_manager import ProbeManager
+
+        self.probe_manager = ProbeManager()
 
     # ------------------------------------------------------------------
     # Informer callbacks
@@ -149,6 +153,17 @@ class HivedScheduler:
         with self.lock:
             return dict(self.health_reports)
 
+    def on_probe_result(self, result: dict) -> dict:
+        """Placement-probe result from a node agent; an unhealthy verdict
+        (busbw below the xGMI floor) marks the probed leaf cells bad."""
+        verdict = self.probe_manager.report(result)
+        if not verdict["healthy"]:
+            node = result.get("node", "")
+            with self.lock:
+                for idx in result.get("leafCellIndices", []):
+                    self.algorithm.set_leaf_cell_healthy(node, int(idx), False)
+        return verdict
+
     def _add_bound_pod(self, pod: dict) -> None:
         """Recovery path: rebuild allocation state from the pod-bind-info
         annotation (the pods ARE the database)."""
@@ -237,6 +252,10 @@ class HivedScheduler:
                     pod=binding_pod, state=POD_BINDING, pod_scheduling_spec=spec,
                     pod_bind_info=info, node=info.node)
                 self.pod_statuses[uid] = st
+                # queue post-bind placement validation (RCCL probe on the node)
+                self.probe_manager.enqueue_group(
+                    spec.affinityGroup.name if spec.affinityGroup else key,
+                    {info.node: list(info.leafCellIsolation)})
                 if _HAVE_PROM:
                     _DECISIONS.labels("bind").inc()
                 if self._should_force_bind(st, suggested):

@@ -103,6 +103,19 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
     async def health_reports():
         return scheduler.get_health_reports()
 
+    # ---- post-bind placement probes (agents poll tasks, post results) ----
+    @app.get("/v1/health/probes/{node}")
+    async def probe_tasks(node: str):
+        return scheduler.probe_manager.poll(node)
+
+    @app.post("/v1/health/probes")
+    async def probe_result(result: Dict[str, Any]):
+        return scheduler.on_probe_result(result)
+
+    @app.get("/v1/inspect/probes/{group:path}")
+    async def probe_results(group: str):
+        return scheduler.probe_manager.group_results(group)
+
     # ---- metrics ----
     @app.get(constants.MetricsPath)
     async def metrics():

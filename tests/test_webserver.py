@@ -101,3 +101,39 @@ def test_health_report_endpoint(client):
     assert rr.status_code == 200 and rr.json().get("NodeNames")
     reports = client.get("/v1/inspect/health").json()
     assert "node1" in reports
+
+
+def test_probe_workflow(client):
+    # binding a pod enqueues a placement-probe task for its node
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    tasks = client.get("/v1/health/probes/node1").json()
+    assert len(tasks) == 1
+    assert tasks[0]["leafCellIndices"] == [0, 1]
+    group = tasks[0]["group"]
+    # polling drains the queue
+    assert client.get("/v1/health/probes/node1").json() == []
+    # a healthy probe result is recorded
+    r = client.post("/v1/health/probes", json={
+        "group": group, "node": "node1", "leafCellIndices": [0, 1],
+        "ok": True, "busbw_gbps": 142.0, "algbw_gbps": 142.0})
+    assert r.json()["healthy"] is True
+    results = client.get(f"/v1/inspect/probes/{group}").json()
+    assert results and results[0]["busbw_gbps"] == 142.0
+
+
+def test_probe_degraded_link_marks_cells_bad(client):
+    pod = make_pod("p1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    task = client.get("/v1/health/probes/node1").json()[0]
+    # degraded xGMI: far below the ~153 GB/s link expectation
+    r = client.post("/v1/health/probes", json={
+        "group": task["group"], "node": "node1", "leafCellIndices": [0, 1],
+        "ok": True, "busbw_gbps": 9.0})
+    assert r.json()["healthy"] is False
+    # the pair is now bad: next pair request avoids GPUs 0/1
+    pod2 = make_pod("p2", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    rr = client.post(constants.FilterPath, json={"Pod": pod2, "NodeNames": ["node1"]})
+    assert rr.json().get("NodeNames")
+    st = client.get(constants.AffinityGroupsPath + "ns/p2").json()
+    assert set(st["physicalPlacement"]["node1"]).isdisjoint({0, 1})
