@@ -881,7 +881,16 @@ std::pair<bool, std::string> HivedCore::allocateLeafCell(PhysicalCell* p, Virtua
 }
 
 void HivedCore::releaseLeafCell(PhysicalCell* p, const std::string& vc) {
-  if (VirtualCell* v = p->virt) {
+  (void)vc;  // the owning VC is derived from the binding, not the releasing
+             // group: an opportunistic pod of VC-A can sit on a leaf carrying
+             // VC-B's bad-cell binding, and must not touch VC-B's accounting
+  if (!p->otVC.empty() || p->virt == nullptr) {
+    // opportunistic allocation: physical side only; a bad-cell binding (if
+    // any) belongs to the doomed-bad machinery and stays
+    p->otVC.clear();
+  } else {
+    VirtualCell* v = p->virt;
+    const std::string& owner = v->vc;
     updateUsedLeafCellNumAtPriority(v, v->priority, false);
     setCellPriority(v, kFreePriority);
     PhysicalCell* preassignedPhysical = v->preassigned->phys;
@@ -891,12 +900,10 @@ void HivedCore::releaseLeafCell(PhysicalCell* p, const std::string& vc) {
     }
     if (preassignedPhysical != nullptr && !preassignedPhysical->pinned &&
         v->preassigned->priority < kMinGuaranteedPriority &&
-        !vcDoomedBadCells_[vc][preassignedPhysical->chain].contains(preassignedPhysical,
-                                                                    preassignedPhysical->level)) {
-      releasePreassignedCell(preassignedPhysical, vc, false);
+        !vcDoomedBadCells_[owner][preassignedPhysical->chain].contains(
+            preassignedPhysical, preassignedPhysical->level)) {
+      releasePreassignedCell(preassignedPhysical, owner, false);
     }
-  } else {
-    p->otVC.clear();
   }
   updateUsedLeafCellNumAtPriority(p, p->priority, false);
   setCellPriority(p, kFreePriority);

@@ -463,6 +463,13 @@ void HivedCore::setHealthyCell(PhysicalCell* c) {
         allVCDoomedBadCellNum_[c->chain][c->level]--;
         releasePreassignedCell(c, vc->vc, true);
       }
+    } else if (!c->pinned && vc->parent == nullptr &&
+               vcDoomedBadCells_[vc->vc][c->chain].contains(c, c->level)) {
+      // a doomed bad cell now healthy AND in real use: it is no longer
+      // "doomed" — it transitions to a normally-allocated preassigned cell
+      // (keeps its binding; releaseLeafCell will release it when free)
+      vcDoomedBadCells_[vc->vc][c->chain].remove(c, c->level);
+      allVCDoomedBadCellNum_[c->chain][c->level]--;
     }
   }
   if (c->parent == nullptr) return;
@@ -521,14 +528,22 @@ void HivedCore::tryUnbindDoomedBadCell(const std::string& chain, int level) {
     if (chainIt == vcFreeNum.end()) continue;
     auto& doomed = vcDoomedBadCells_[vcName][chain];
     if (level > doomed.top()) continue;
-    while (!doomed.at(level).empty() &&
-           chainIt->second[level] <
-               totalLeftCellNum_[chain][level] -
-                   static_cast<int>(badFreeCells_[chain].at(level).size())) {
-      auto* pc = static_cast<PhysicalCell*>(vcDoomedBadCells_[vcName][chain].at(level)[0]);
+    while (chainIt->second[level] <
+           totalLeftCellNum_[chain][level] -
+               static_cast<int>(badFreeCells_[chain].at(level).size())) {
+      // release only doomed cells whose binding exists purely because they
+      // are bad (not in real use by a group scheduled onto healthy leaves)
+      PhysicalCell* pc = nullptr;
+      for (Cell* cand : doomed.at(level)) {
+        if (cand->priority < kMinGuaranteedPriority) {
+          pc = static_cast<PhysicalCell*>(cand);
+          break;
+        }
+      }
+      if (pc == nullptr) break;
       pc->virt->phys = nullptr;
       pc->virt = nullptr;
-      vcDoomedBadCells_[vcName][chain].remove(pc, level);
+      doomed.remove(pc, level);
       allVCDoomedBadCellNum_[chain][level]--;
       releasePreassignedCell(pc, vcName, true);
     }

@@ -416,6 +416,7 @@ class PyHivedCore {
   }
 
   long long scheduleCount() const { return core_.scheduleCount_; }
+  void checkInvariants() const { hived::checkInvariants(core_); }
 
  private:
   HivedCore core_;
@@ -459,5 +460,6 @@ PYBIND11_MODULE(hivedcore, m) {
       .def("get_physical_cluster_status", &PyHivedCore::getPhysicalClusterStatus)
       .def("get_all_virtual_clusters_status", &PyHivedCore::getAllVirtualClustersStatus)
       .def("get_virtual_cluster_status", &PyHivedCore::getVirtualClusterStatus, py::arg("vc"))
-      .def("schedule_count", &PyHivedCore::scheduleCount);
+      .def("schedule_count", &PyHivedCore::scheduleCount)
+      .def("check_invariants", &PyHivedCore::checkInvariants);
 }

@@ -19,6 +19,7 @@ SOURCES = [
     "build.cpp",
     "alloc.cpp",
     "algorithm.cpp",
+    "debug.cpp",
     "bindings.cpp",
 ]
 

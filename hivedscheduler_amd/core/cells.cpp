@@ -63,7 +63,7 @@ void bindCell(PhysicalCell* pc, VirtualCell* vc) {
 // sibling remains; never unbind pinned cells.
 void unbindCell(PhysicalCell* c) {
   VirtualCell* boundVirtual = c->virt;
-  while (!boundVirtual->phys->pinned) {
+  while (boundVirtual->phys != nullptr && !boundVirtual->phys->pinned) {
     PhysicalCell* boundPhysical = boundVirtual->phys;
     boundVirtual->phys = nullptr;
     boundPhysical->virt = nullptr;

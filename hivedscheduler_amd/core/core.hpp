@@ -473,5 +473,6 @@ PhysicalCell* findPhysicalLeafCell(std::map<std::string, ChainCellList>& fullCel
                                    const std::string& chain, const std::string& node,
                                    int leafIndex);
 Cell* ancestorNoHigherThanNode(Cell* c);
+void checkInvariants(const HivedCore& core);
 
 }  // namespace hived
