@@ -21,11 +21,13 @@ def get_ops():
     return _ops
 
 
-def gpu_health_report(device: int = 0, quick: bool = True) -> dict:
+def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False) -> dict:
     """Run the health-probe kernels on one GPU and return measured facts.
 
-    Feeds leaf-cell healthiness: HBM bandwidth deficit or MFMA mismatch marks
-    the GPU's leaf cell bad.
+    Feeds leaf-cell healthiness: HBM bandwidth deficit, MFMA mismatch, or
+    (deep=True) any stuck-bit error in the HBM pattern sweep marks the GPU's
+    leaf cell bad. The deep sweep scans a bounded span (16 GiB quick / 64 GiB
+    otherwise) of the 288 GB HBM3E in 4 GiB chunks.
     """
     import torch
 
@@ -47,4 +49,8 @@ def gpu_health_report(device: int = 0, quick: bool = True) -> dict:
     report["mfma_cross_cu_spread"] = tile_spread  # must be exactly 0
     report["mfma_ok"] = bool(tile_spread == 0.0 and max_err < 0.1)
     report["healthy"] = bool(report["mfma_ok"] and report["hbm_gbps"] > 1000.0)
+    if deep:
+        sweep = ops.hbm_sweep(16 if quick else 64, 4, 1)
+        report["hbm_sweep"] = sweep
+        report["healthy"] = bool(report["healthy"] and sweep["errors"] == 0)
     return report

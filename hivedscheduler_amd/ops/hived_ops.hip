@@ -185,6 +185,92 @@ double p2p_gbps(int src_dev, int dst_dev, long size_mb, long iters) {
   return (double)bytes * iters / sec / 1e9;
 }
 
+// ---------------------------------------------------------------------------
+// HBM pattern sweep: stuck-bit / corruption scan over a bounded span of HBM.
+// Writes an address-derived 64-bit pattern, reads it back after the full
+// write pass (chunk >> L2, so read-back hits HBM), counts mismatches with a
+// device-side atomic. Catches data-integrity faults that bandwidth tests
+// (triad) cannot see. Memory-bounded: one chunk allocated at a time with
+// plain hipMalloc; allocation failure ends the sweep gracefully.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned long long mix64(unsigned long long x) {
+  // splitmix64 finalizer: full-period, all 64 bits participate
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__global__ void pattern_write_kernel(unsigned long long* __restrict__ p, long n,
+                                     unsigned long long seed) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) p[i] = mix64(seed ^ (unsigned long long)i);
+}
+
+__global__ void pattern_verify_kernel(const unsigned long long* __restrict__ p, long n,
+                                      unsigned long long seed,
+                                      unsigned long long* __restrict__ errors) {
+  long i = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  unsigned long long local = 0;
+  for (; i < n; i += stride)
+    if (p[i] != mix64(seed ^ (unsigned long long)i)) local++;
+  if (local) atomicAdd(errors, local);
+}
+
+// Sweep up to max_gib GiB of this device's HBM in chunk_gib chunks.
+// Returns {bytes_tested, errors, write_gbps, verify_gbps}.
+py::dict hbm_sweep(long max_gib, long chunk_gib, long seed) {
+  TORCH_CHECK(max_gib > 0 && chunk_gib > 0 && chunk_gib <= max_gib);
+  const long chunk_bytes = chunk_gib << 30;
+  const long n = chunk_bytes / 8;
+  unsigned long long* errs_d = nullptr;
+  HIP_CHECK(hipMalloc(&errs_d, sizeof(unsigned long long)));
+  HIP_CHECK(hipMemset(errs_d, 0, sizeof(unsigned long long)));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipEvent_t start, stop;
+  HIP_CHECK(hipEventCreate(&start));
+  HIP_CHECK(hipEventCreate(&stop));
+  double write_s = 0, verify_s = 0;
+  long tested = 0;
+  int blocks = 8192, threads = 256;  // >> 256 CUs, fills all 8 XCDs
+  for (long off = 0; off + chunk_bytes <= (max_gib << 30); off += chunk_bytes) {
+    unsigned long long* p = nullptr;
+    // keep headroom: stop if the allocator can't give another chunk
+    if (hipMalloc(&p, chunk_bytes) != hipSuccess) { (void)hipGetLastError(); break; }
+    unsigned long long chunk_seed = (unsigned long long)seed ^ (unsigned long long)off;
+    float ms = 0;
+    HIP_CHECK(hipEventRecord(start, stream));
+    hipLaunchKernelGGL(pattern_write_kernel, dim3(blocks), dim3(threads), 0, stream, p, n,
+                       chunk_seed);
+    HIP_CHECK(hipEventRecord(stop, stream));
+    HIP_CHECK(hipEventSynchronize(stop));
+    HIP_CHECK(hipEventElapsedTime(&ms, start, stop));
+    write_s += ms / 1e3;
+    HIP_CHECK(hipEventRecord(start, stream));
+    hipLaunchKernelGGL(pattern_verify_kernel, dim3(blocks), dim3(threads), 0, stream, p, n,
+                       chunk_seed, errs_d);
+    HIP_CHECK(hipEventRecord(stop, stream));
+    HIP_CHECK(hipEventSynchronize(stop));
+    HIP_CHECK(hipEventElapsedTime(&ms, start, stop));
+    verify_s += ms / 1e3;
+    HIP_CHECK(hipFree(p));
+    tested += chunk_bytes;
+  }
+  unsigned long long errs = 0;
+  HIP_CHECK(hipMemcpy(&errs, errs_d, sizeof(errs), hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(errs_d));
+  HIP_CHECK(hipEventDestroy(start));
+  HIP_CHECK(hipEventDestroy(stop));
+  py::dict d;
+  d["bytes_tested"] = (long long)tested;
+  d["errors"] = (long long)errs;
+  d["write_gbps"] = write_s > 0 ? tested / write_s / 1e9 : 0.0;
+  d["verify_gbps"] = verify_s > 0 ? tested / verify_s / 1e9 : 0.0;
+  return d;
+}
+
 py::dict device_info(int dev) {
   hipDeviceProp_t prop;
   HIP_CHECK(hipGetDeviceProperties(&prop, dev));
@@ -213,6 +299,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("repeats") = 1, "Per-wave bf16 MFMA tile GEMM across all CUs");
   m.def("p2p_gbps", &p2p_gbps, py::arg("src_dev"), py::arg("dst_dev"), py::arg("size_mb") = 256,
         py::arg("iters") = 10, "xGMI peer-to-peer copy bandwidth in GB/s");
+  m.def("hbm_sweep", &hbm_sweep, py::arg("max_gib") = 16, py::arg("chunk_gib") = 4,
+        py::arg("seed") = 1,
+        "Stuck-bit pattern sweep over up to max_gib GiB of HBM; returns error count");
   m.def("device_info", &device_info, py::arg("dev") = 0);
   m.def("device_count", &device_count);
 }

@@ -118,3 +118,25 @@ def test_bench_on_gpu():
     result = json.loads(out.stdout.strip().splitlines()[-1])
     assert result["config"]["vc_safety_violations"] == 0
     assert result["config"]["rccl_busbw_gbps"], "expected a busbw measurement on GPU"
+
+
+@needs_gpu
+def test_hbm_pattern_sweep():
+    """Stuck-bit sweep over 8 GiB: zero errors on healthy HBM, bandwidth in
+    the HBM class (not cache-resident: chunk 4 GiB >> 512 MB Infinity Cache)."""
+    from hivedscheduler_amd.ops import get_ops
+
+    ops = get_ops()
+    r = ops.hbm_sweep(8, 4, 12345)
+    assert r["bytes_tested"] == 8 << 30
+    assert r["errors"] == 0, f"HBM corruption detected: {r}"
+    assert r["write_gbps"] > 500 and r["verify_gbps"] > 500, r
+
+
+@needs_gpu
+def test_health_report_deep():
+    from hivedscheduler_amd.ops import gpu_health_report
+
+    rep = gpu_health_report(0, quick=True, deep=True)
+    assert rep["hbm_sweep"]["errors"] == 0
+    assert rep["healthy"]
