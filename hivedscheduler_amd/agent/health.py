@@ -47,9 +47,12 @@ def collect_node_health(
     min_hbm_gbps: float = 2000.0,
     min_pair_busbw_gbps: float = 50.0,
     probe_pairs: bool = False,
+    sweep: bool = False,
 ) -> dict:
     """One health sweep over all visible GPUs. deep=True also runs the HIP
-    kernels (HBM + MFMA); probe_pairs=True RCCL-probes each xGMI pair."""
+    kernels (HBM + MFMA); sweep=True adds the 16 GiB HBM stuck-bit pattern
+    sweep (any error => leaf bad); probe_pairs=True RCCL-probes each xGMI
+    pair."""
     report: dict = {"node": socket.gethostname(), "time": time.time(), "gpus": {}}
     alive = _rocm_smi_gpu_state()
     try:
@@ -67,12 +70,14 @@ def collect_node_health(
 
         for i in range(n):
             try:
-                rep = gpu_health_report(i, quick=True)
+                rep = gpu_health_report(i, quick=True, deep=sweep)
                 report["gpus"][str(i)].update(
                     hbm_gbps=round(rep["hbm_gbps"], 1),
                     mfma_ok=rep["mfma_ok"],
                 )
-                if not rep["mfma_ok"] or rep["hbm_gbps"] < min_hbm_gbps:
+                if "hbm_sweep" in rep:
+                    report["gpus"][str(i)]["hbm_sweep_errors"] = rep["hbm_sweep"]["errors"]
+                if not rep["healthy"] or rep["hbm_gbps"] < min_hbm_gbps:
                     report["gpus"][str(i)]["healthy"] = False
             except Exception as e:
                 report["gpus"][str(i)].update(healthy=False, error=str(e)[:200])
@@ -102,11 +107,13 @@ class NodeHealthAgent:
 
     def __init__(self, scheduler_url: str, node_name: Optional[str] = None,
                  interval_s: float = 60.0, deep_every: int = 10,
-                 probe_pairs: bool = False):
+                 sweep_every: int = 60, probe_pairs: bool = False):
         self.scheduler_url = scheduler_url.rstrip("/")
         self.node_name = node_name or socket.gethostname()
         self.interval_s = interval_s
         self.deep_every = deep_every
+        # stuck-bit HBM sweep cadence (GPU must be idle-ish; 16 GiB ~ 5 s/GPU)
+        self.sweep_every = sweep_every
         self.probe_pairs = probe_pairs
         self._rounds = 0
 
@@ -153,8 +160,10 @@ class NodeHealthAgent:
 
     def run_once(self) -> dict:
         deep = (self._rounds % self.deep_every) == 0
+        sweep = (self._rounds % self.sweep_every) == 0
         self._rounds += 1
-        report = collect_node_health(deep=deep, probe_pairs=self.probe_pairs and deep)
+        report = collect_node_health(deep=deep, probe_pairs=self.probe_pairs and deep,
+                                     sweep=sweep and deep)
         report["node"] = self.node_name
         self.post_report(report)
         report["placement_probes"] = self.run_placement_probes()

