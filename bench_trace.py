@@ -1,0 +1,215 @@
+#!/usr/bin/env python3
+"""OSDI'20-style trace replay: a virtual-time discrete-event simulation of a
+multi-tenant cluster under churn, driving the C++ scheduling core through the
+exact extender protocol (filter -> optimistic commit; preempt -> victim
+delete -> refilter; delete on completion).
+
+The reference publishes its quantitative evaluation only in the OSDI'20 paper
+(BASELINE.md); this replay reproduces that style of workload — Poisson
+arrivals, mixed 1/2/4/8-GPU jobs, three VCs, guaranteed + opportunistic
+priorities, gang groups — and reports per-decision latency percentiles,
+queueing delay, preemption counts, utilization, and VC-safety (checked with
+the core's full invariant checker every N events, plus an end-of-trace
+whole-quota drain test).
+
+Usage: python bench_trace.py [--jobs 2000] [--nodes 4] [--seed 0] [--out f.json]
+"""
+import argparse
+import heapq
+import json
+import random
+import statistics
+import time
+
+
+def pcts(samples):
+    if not samples:
+        return {}
+    s = sorted(samples)
+    return {"p50": round(s[len(s) // 2], 6),
+            "p95": round(s[min(len(s) - 1, int(len(s) * 0.95))], 6),
+            "p99": round(s[min(len(s) - 1, int(len(s) * 0.99))], 6),
+            "mean": round(statistics.fmean(s), 6), "n": len(s)}
+
+
+class TraceReplay:
+    def __init__(self, nodes=4, seed=0, invariant_every=200, arrival_mean_s=12.0):
+        from hivedscheduler_amd.sim import SimScheduler, mi355x_cluster_config
+
+        half, rest = nodes // 2, nodes - nodes // 2 - nodes // 4
+        self.sim = SimScheduler(mi355x_cluster_config(num_nodes=nodes, vcs={
+            "prod": [("MI355X-NODE", half)],
+            "research": [("MI355X-NODE", nodes // 4),
+                         ("MI355X-NODE.MI355X-QUAD", rest)],
+            "dev": [("MI355X-NODE.MI355X-QUAD", rest)],
+        }))
+        self.rng = random.Random(seed)
+        self.nodes = nodes
+        self.events = []  # (t, seq, kind, payload)
+        self.seq = 0
+        self.now = 0.0
+        self.decision_lat_ms = []
+        self.queue_delay = []  # virtual-time arrival -> bind
+        self.stats = {"binds": 0, "waits": 0, "preemptions": 0, "completions": 0,
+                      "victim_resubmits": 0, "invariant_checks": 0}
+        self.invariant_every = invariant_every
+        self.arrival_mean_s = arrival_mean_s
+        self.decisions = 0
+        self.gpu_time_used = 0.0  # GPU-seconds of completed work
+        self.jobs = {}  # key -> job dict
+
+    def push(self, t, kind, payload):
+        self.seq += 1
+        heapq.heappush(self.events, (t, self.seq, kind, payload))
+
+    def new_job(self, i):
+        vc = self.rng.choice(["prod", "prod", "research", "research", "dev"])
+        prio = self.rng.choice([-1, -1, 0, 0, 0, 1, 10])
+        size = self.rng.choice([1, 1, 1, 2, 2, 4, 4, 8])
+        gang = size == 8 and self.rng.random() < 0.5
+        dur = self.rng.expovariate(1 / 120.0) + 10  # mean ~130 s
+        return {"key": f"t/j{i}", "vc": vc, "priority": prio, "size": size,
+                "gang": gang, "duration": dur, "arrival": None, "pods": []}
+
+    def spec_for(self, job):
+        if job["gang"]:  # 2 pods x 4 GPUs, one affinity group
+            return self.sim.pod_spec(vc=job["vc"], priority=job["priority"],
+                                     leaf_cells=4, group=job["key"],
+                                     members=[(2, 4)])
+        return self.sim.pod_spec(vc=job["vc"], priority=job["priority"],
+                                 leaf_cells=job["size"])
+
+    def pod_keys(self, job):
+        return [f"{job['key']}/p{k}" for k in range(2)] if job["gang"] else [job["key"]]
+
+    def try_schedule(self, job):
+        """One scheduling attempt for all pods of the job. Returns bound?"""
+        spec = self.spec_for(job)
+        keys = self.pod_keys(job)
+        results = []
+        for k in keys:
+            t0 = time.perf_counter_ns()
+            r = self.sim.schedule(k, spec)
+            self.decision_lat_ms.append((time.perf_counter_ns() - t0) / 1e6)
+            self.decisions += 1
+            results.append((k, r))
+            if r.kind != "bind":
+                break
+        if all(r.kind == "bind" for _, r in results) and len(results) == len(keys):
+            job["pods"] = keys
+            self.stats["binds"] += 1
+            self.queue_delay.append(self.now - job["arrival"])
+            self.push(self.now + job["duration"], "finish", job)
+            return True
+        # roll back partial gang binds (K8s would retry the group next cycle)
+        for k, r in results:
+            if r.kind == "bind":
+                self.sim.delete_pod(k)
+        last = results[-1][1]
+        if last.kind == "preempt" and job["priority"] >= 0:
+            # run the K8s preemption protocol: delete victims, resubmit them
+            final = self.sim.run_preemption_to_completion(keys[0], spec)
+            self.stats["preemptions"] += 1
+            victims = [j for j in self.jobs.values()
+                       if j["pods"] and any(p not in self.sim.pods for p in j["pods"])]
+            for v in victims:
+                v["pods"] = []
+                self.stats["victim_resubmits"] += 1
+                self.push(self.now + 1.0, "retry", v)
+            if final.kind == "bind":
+                ok = True
+                for k in keys[1:]:
+                    if self.sim.schedule(k, spec).kind != "bind":
+                        ok = False
+                        break
+                if ok:
+                    job["pods"] = keys
+                    self.stats["binds"] += 1
+                    self.queue_delay.append(self.now - job["arrival"])
+                    self.push(self.now + job["duration"], "finish", job)
+                    return True
+                for k in keys:
+                    if k in self.sim.pods:
+                        self.sim.delete_pod(k)
+        self.stats["waits"] += 1
+        self.push(self.now + 5.0, "retry", job)  # default-scheduler retry cadence
+        return False
+
+    def run(self, n_jobs):
+        t = 0.0
+        for i in range(n_jobs):
+            t += self.rng.expovariate(1 / self.arrival_mean_s)
+            job = self.new_job(i)
+            self.jobs[job["key"]] = job
+            self.push(t, "arrive", job)
+        horizon = t + 3600.0
+        check = self.sim.alg._core.check_invariants
+        while self.events:
+            self.now, _, kind, job = heapq.heappop(self.events)
+            if self.now > horizon:
+                break
+            if kind == "arrive":
+                job["arrival"] = self.now
+                self.try_schedule(job)
+            elif kind == "retry":
+                if job["pods"] or job["key"] not in self.jobs:
+                    continue  # bound meanwhile or gone
+                self.try_schedule(job)
+            elif kind == "finish":
+                if not job["pods"]:
+                    continue  # was victimized; a retry event exists
+                for k in job["pods"]:
+                    if k in self.sim.pods:
+                        self.sim.delete_pod(k)
+                self.gpu_time_used += job["size"] * job["duration"]
+                job["pods"] = []
+                del self.jobs[job["key"]]
+                self.stats["completions"] += 1
+            if self.decisions and self.decisions % self.invariant_every == 0:
+                check()
+                self.stats["invariant_checks"] += 1
+        # end-of-trace drain + VC-safety: every VC can take its full quota
+        for k in list(self.sim.pods):
+            self.sim.delete_pod(k)
+        violations = 0
+        n = self.nodes
+        quota = {"prod": [8] * (n // 2), "research": [8] * (n // 4) + [4] * (n - n // 2 - n // 4),
+                 "dev": [4] * (n - n // 2 - n // 4)}
+        for vc, sizes in quota.items():
+            for j, cells in enumerate(sizes):
+                r = self.sim.schedule(f"drain/{vc}{j}",
+                                      self.sim.pod_spec(vc=vc, priority=0, leaf_cells=cells))
+                violations += r.kind != "bind"
+        check()
+        return {
+            "trace": {"jobs": n_jobs, "nodes": self.nodes, "virtual_seconds": round(self.now, 1)},
+            "decision_latency_ms": pcts(self.decision_lat_ms),
+            "queue_delay_virtual_s": pcts(self.queue_delay),
+            "vc_safety_violations": violations,
+            "utilization": round(self.gpu_time_used / (8 * self.nodes * max(self.now, 1)), 3),
+            **self.stats,
+        }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--jobs", type=int, default=2000)
+    ap.add_argument("--nodes", type=int, default=4)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--arrival-mean-s", type=float, default=12.0,
+                    help="mean inter-arrival (12 s ~ 87%% offered load on 4 nodes)")
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+    t0 = time.perf_counter()
+    rep = TraceReplay(nodes=args.nodes, seed=args.seed,
+                      arrival_mean_s=args.arrival_mean_s).run(args.jobs)
+    rep["wall_s"] = round(time.perf_counter() - t0, 2)
+    js = json.dumps(rep, indent=1)
+    print(js)
+    if args.out:
+        with open(args.out, "w") as f:
+            f.write(js + "\n")
+
+
+if __name__ == "__main__":
+    main()

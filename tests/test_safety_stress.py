@@ -117,3 +117,14 @@ def test_schedule_latency_scales_to_large_cluster():
         lat.append((time.perf_counter_ns() - t0) / 1e6)
     lat.sort()
     assert lat[len(lat) // 2] < 5.0, f"p50 regressed: {lat[len(lat)//2]:.3f} ms"
+
+
+def test_trace_replay_small():
+    """100-job OSDI'20-style trace replay: zero VC-safety violations, all
+    invariant checks pass, every job eventually completes or is queued."""
+    from bench_trace import TraceReplay
+
+    rep = TraceReplay(nodes=4, seed=7, invariant_every=100).run(100)
+    assert rep["vc_safety_violations"] == 0
+    assert rep["invariant_checks"] > 0
+    assert rep["binds"] > 0 and rep["completions"] > 0
