@@ -182,6 +182,47 @@ def config5(iters, gpu):
     return out
 
 
+def config_http(iters):
+    """End-to-end extender HTTP latency: JSON ExtenderArgs through the ASGI
+    app (the path kube-scheduler exercises; reference deploy sets
+    httpTimeout: 5s — we measure well under a millisecond)."""
+    import yaml
+    from starlette.testclient import TestClient
+
+    from hivedscheduler_amd.api import constants
+    from hivedscheduler_amd.scheduler import HivedScheduler
+    from hivedscheduler_amd.sim import mi355x_cluster_config
+    from hivedscheduler_amd.webserver import create_app
+
+    sched = HivedScheduler(mi355x_cluster_config(num_nodes=1))
+    sched.on_node_add({
+        "metadata": {"name": "node1", "uid": "node-node1"},
+        "spec": {},
+        "status": {"conditions": [{"type": "Ready", "status": "True"}]},
+    })
+    client = TestClient(create_app(sched), raise_server_exceptions=False)
+    spec = yaml.safe_dump({"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    lat = []
+    for i in range(-5, iters):
+        pod = {
+            "metadata": {"name": f"hp{i}", "namespace": "b", "uid": f"uid-b-hp{i}",
+                         "annotations": {constants.AnnotationKeyPodSchedulingSpec: spec}},
+            "spec": {"containers": [{"resources": {
+                "limits": {constants.ResourceNamePodSchedulingEnable: 1}}}]},
+            "status": {"phase": "Pending"},
+        }
+        r, ms = timed(lambda: client.post(
+            constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]}))
+        assert r.status_code == 200 and r.json().get("NodeNames") == ["node1"], r.text
+        if i >= 0:
+            lat.append(ms)
+        # release so the 8-GPU node never fills (full pod: delete path needs
+        # the spec annotation + opt-in resource limit)
+        sched.on_pod_delete(pod)
+    return {"desc": "extender filter over HTTP (ASGI, JSON in/out)",
+            "http_latency": pcts(lat)}
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=200)
@@ -201,7 +242,8 @@ def main():
                      ("config2", lambda: config2(args.iters, gpu)),
                      ("config3", lambda: config3(args.iters)),
                      ("config4", lambda: config4(args.iters)),
-                     ("config5", lambda: config5(args.iters, gpu))):
+                     ("config5", lambda: config5(args.iters, gpu)),
+                     ("extender_http", lambda: config_http(args.iters))):
         t0 = time.perf_counter()
         report[name] = fn()
         report[name]["wall_s"] = round(time.perf_counter() - t0, 3)
