@@ -183,11 +183,16 @@ def config5(iters, gpu):
 
 
 def config_http(iters):
-    """End-to-end extender HTTP latency: JSON ExtenderArgs through the ASGI
-    app (the path kube-scheduler exercises; reference deploy sets
-    httpTimeout: 5s — we measure well under a millisecond)."""
+    """End-to-end extender HTTP latency: JSON ExtenderArgs over TCP to a real
+    uvicorn server with a keep-alive connection — the path kube-scheduler
+    exercises (reference deploy sets httpTimeout: 5s)."""
+    import http.client
+    import json as jsonlib
+    import socket
+    import threading
+
+    import uvicorn
     import yaml
-    from starlette.testclient import TestClient
 
     from hivedscheduler_amd.api import constants
     from hivedscheduler_amd.scheduler import HivedScheduler
@@ -200,26 +205,51 @@ def config_http(iters):
         "spec": {},
         "status": {"conditions": [{"type": "Ready", "status": "True"}]},
     })
-    client = TestClient(create_app(sched), raise_server_exceptions=False)
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    server = uvicorn.Server(uvicorn.Config(create_app(sched), host="127.0.0.1",
+                                           port=port, log_level="error"))
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    for _ in range(100):
+        if server.started:
+            break
+        time.sleep(0.05)
+    assert server.started, "uvicorn did not start"
+
+    conn = http.client.HTTPConnection("127.0.0.1", port)
     spec = yaml.safe_dump({"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
     lat = []
-    for i in range(-5, iters):
-        pod = {
-            "metadata": {"name": f"hp{i}", "namespace": "b", "uid": f"uid-b-hp{i}",
-                         "annotations": {constants.AnnotationKeyPodSchedulingSpec: spec}},
-            "spec": {"containers": [{"resources": {
-                "limits": {constants.ResourceNamePodSchedulingEnable: 1}}}]},
-            "status": {"phase": "Pending"},
-        }
-        r, ms = timed(lambda: client.post(
-            constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]}))
-        assert r.status_code == 200 and r.json().get("NodeNames") == ["node1"], r.text
-        if i >= 0:
-            lat.append(ms)
-        # release so the 8-GPU node never fills (full pod: delete path needs
-        # the spec annotation + opt-in resource limit)
-        sched.on_pod_delete(pod)
-    return {"desc": "extender filter over HTTP (ASGI, JSON in/out)",
+    try:
+        for i in range(-10, iters):
+            pod = {
+                "metadata": {"name": f"hp{i}", "namespace": "b", "uid": f"uid-b-hp{i}",
+                             "annotations": {constants.AnnotationKeyPodSchedulingSpec: spec}},
+                "spec": {"containers": [{"resources": {
+                    "limits": {constants.ResourceNamePodSchedulingEnable: 1}}}]},
+                "status": {"phase": "Pending"},
+            }
+            body = jsonlib.dumps({"Pod": pod, "NodeNames": ["node1"]})
+
+            def post():
+                conn.request("POST", constants.FilterPath, body,
+                             {"Content-Type": "application/json"})
+                resp = conn.getresponse()
+                return resp.status, resp.read()
+
+            (status, data), ms = timed(post)
+            assert status == 200 and jsonlib.loads(data).get("NodeNames") == ["node1"], data
+            if i >= 0:
+                lat.append(ms)
+            # release so the 8-GPU node never fills (full pod: delete path
+            # needs the spec annotation + opt-in resource limit)
+            sched.on_pod_delete(pod)
+    finally:
+        conn.close()
+        server.should_exit = True
+        th.join(timeout=5)
+    return {"desc": "extender filter over HTTP (uvicorn, keep-alive, JSON in/out)",
             "http_latency": pcts(lat)}
 
 

@@ -45,14 +45,27 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
         return PlainTextResponse("ok")
 
     # ---- extender verbs ----
+    async def _json_body(request: Request) -> Dict[str, Any]:
+        # raw json parse: pydantic body validation costs ~1 ms per call on
+        # the filter hot path and the extender args are plain dicts anyway
+        try:
+            args = await request.json()
+        except Exception:
+            raise WebServerError.bad_request("request body is not valid JSON")
+        if not isinstance(args, dict):
+            raise WebServerError.bad_request("request body must be a JSON object")
+        return args
+
     @app.post(constants.FilterPath)
-    async def filter_verb(args: Dict[str, Any]):
+    async def filter_verb(request: Request):
+        args = await _json_body(request)
         if not isinstance(args.get("Pod"), dict):
             raise WebServerError.bad_request("ExtenderArgs.Pod is missing")
-        return scheduler.filter(args)
+        return JSONResponse(scheduler.filter(args))
 
     @app.post(constants.BindPath)
-    async def bind_verb(args: Dict[str, Any]):
+    async def bind_verb(request: Request):
+        args = await _json_body(request)
         for field in ("PodName", "PodNamespace", "PodUID", "Node"):
             if not args.get(field):
                 raise WebServerError.bad_request(f"ExtenderBindingArgs.{field} is missing")
@@ -64,10 +77,11 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
             return {"Error": e.message}
 
     @app.post(constants.PreemptPath)
-    async def preempt_verb(args: Dict[str, Any]):
+    async def preempt_verb(request: Request):
+        args = await _json_body(request)
         if not isinstance(args.get("Pod"), dict):
             raise WebServerError.bad_request("ExtenderPreemptionArgs.Pod is missing")
-        return scheduler.preempt(args)
+        return JSONResponse(scheduler.preempt(args))
 
     # ---- inspect API ----
     @app.get(constants.AffinityGroupsPath)
