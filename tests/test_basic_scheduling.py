@@ -151,3 +151,22 @@ def test_guaranteed_within_quota_always_schedulable(design_sim):
         assert r.kind == "bind", f"VC1 request {i} ({cells} cells) failed: {r}"
     r = sim.schedule("ns/v1-pin", sim.pod_spec(vc="VC1", leaf_cells=8, pinned_cell_id="VC1-PIN"))
     assert r.kind == "bind"
+
+
+def test_placement_determinism():
+    """Identical request sequences produce identical placements across two
+    fresh schedulers (no randomness in the decision path — important for
+    debugging and for the reference's golden-expectation test style)."""
+    from hivedscheduler_amd.sim import SimScheduler, mi355x_cluster_config
+
+    def run():
+        sim = SimScheduler(mi355x_cluster_config(
+            num_nodes=2, vcs={"VC1": [("MI355X-NODE", 2)]}))
+        out = []
+        for i, cells in enumerate([1, 2, 4, 1, 8, 2, 1]):
+            r = sim.schedule(f"d/p{i}", sim.pod_spec(vc="VC1", leaf_cells=cells))
+            out.append((r.kind, r.bind_info.node if r.kind == "bind" else "",
+                        tuple(sorted(r.bind_info.leafCellIsolation)) if r.kind == "bind" else ()))
+        return out
+
+    assert run() == run()

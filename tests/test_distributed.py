@@ -64,3 +64,32 @@ def test_bench_single_process():
     assert out.returncode == 0, out.stderr[-2000:]
     result = json.loads(out.stdout.strip().splitlines()[-1])
     assert result["config"]["vc_safety_violations"] == 0
+
+
+def test_cell_probe_subgroups_gloo_world2(tmp_path):
+    """Exercises bench.py's rccl_cell_probes subgroup logic (new_group for
+    size<world, probe inside, barrier outside) on gloo — the exact code the
+    driver's unattended 8-GPU run executes over RCCL."""
+    script = tmp_path / "subgroup_main.py"
+    script.write_text(
+        """
+import os, sys, json
+sys.path.insert(0, %r)
+import torch.distributed as dist
+dist.init_process_group(backend="gloo")
+import bench
+world = dist.get_world_size()
+rank = dist.get_rank()
+res = bench.rccl_cell_probes(world, rank, rank)
+if rank == 0:
+    print("SUBGROUP_RESULT " + json.dumps(res))
+dist.destroy_process_group()
+"""
+        % REPO
+    )
+    out = run_torchrun([str(script)])
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("SUBGROUP_RESULT")][0]
+    res = json.loads(line.split(" ", 1)[1])
+    assert set(res) == {"1", "2"}, res
+    assert all(v > 0 for v in res.values()), res
