@@ -28,3 +28,27 @@ def test_standalone_entry_config_parses():
     cfg = new_config(yaml.safe_load(cm["data"]["hivedscheduler.yaml"]))
     alg = HivedAlgorithm(cfg)
     assert alg.all_nodes() == ["mi355x-node-1"]
+
+
+def test_example_request_pods_parse_and_validate():
+    """Every example request pod YAML parses into a valid scheduling spec —
+    including the unmodified legacy-HiveD pod."""
+    import glob
+
+    import yaml
+
+    from hivedscheduler_amd.internal import pod as podmod
+
+    files = sorted(glob.glob(os.path.join(REPO, "examples/request/*.yaml")))
+    assert len(files) >= 4
+    for f in files:
+        p = yaml.safe_load(open(f))
+        assert podmod.is_hived_enabled(p), f
+        spec = podmod.extract_pod_scheduling_spec(p)
+        assert spec.virtualCluster and spec.leafCellNumber > 0, f
+        if "legacy" in f:
+            assert spec.leafCellType == "MI355X" and spec.leafCellNumber == 4
+        if "pinned" in f:
+            assert spec.pinnedCellId
+        if "opportunistic" in f:
+            assert spec.priority == -1
