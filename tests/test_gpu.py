@@ -140,3 +140,70 @@ def test_health_report_deep():
     rep = gpu_health_report(0, quick=True, deep=True)
     assert rep["hbm_sweep"]["errors"] == 0
     assert rep["healthy"]
+
+
+@needs_gpu
+def test_agent_end_to_end_on_hardware():
+    """Hardware-in-the-loop: the node agent collects REAL GPU health (HIP
+    kernels incl. the stuck-bit sweep) over the live uvicorn extender server;
+    the scheduler applies it to the cell tree and serves it via inspect."""
+    import socket
+    import threading
+    import time as _time
+
+    import requests
+    import uvicorn
+
+    from hivedscheduler_amd.agent.health import NodeHealthAgent, collect_node_health
+    from hivedscheduler_amd.scheduler import HivedScheduler
+    from hivedscheduler_amd.sim import mi355x_cluster_config
+    from hivedscheduler_amd.webserver import create_app
+
+    node = socket.gethostname()
+    # simulated view of this box as node "<hostname>" (1 node x 8 GPUs)
+    sched = HivedScheduler(mi355x_cluster_config(num_nodes=1, node_prefix=""))
+    # rename: config names the node ""+"1"; map hostname via health intake only
+    first_node = sched.algorithm.all_nodes()[0]
+    sched.on_node_add({"metadata": {"name": first_node, "uid": "n1"}, "spec": {},
+                       "status": {"conditions": [{"type": "Ready", "status": "True"}]}})
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    server = uvicorn.Server(uvicorn.Config(create_app(sched), host="127.0.0.1",
+                                           port=port, log_level="error"))
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    for _ in range(100):
+        if server.started:
+            break
+        _time.sleep(0.05)
+    try:
+        # real kernels: deep + stuck-bit sweep (bounded 16 GiB)
+        report = collect_node_health(deep=True, sweep=True)
+        assert report["gpus"], report
+        g0 = report["gpus"]["0"]
+        assert g0["mfma_ok"] and g0["hbm_gbps"] > 2000, g0
+        assert g0.get("hbm_sweep_errors", 0) == 0, g0
+        # post as the agent does, against the node name the scheduler knows
+        agent = NodeHealthAgent(f"http://127.0.0.1:{port}", node_name=first_node)
+        assert agent.post_report(report)
+        served = requests.get(f"http://127.0.0.1:{port}/v1/inspect/health",
+                              timeout=10).json()
+        assert first_node in served
+        assert served[first_node]["gpus"]["0"]["mfma_ok"] is True
+        # a degraded-GPU report marks the leaf bad in the live cell tree
+        bad = {"gpus": {"5": {"healthy": False, "mfma_ok": False}}}
+        r = requests.post(f"http://127.0.0.1:{port}/v1/health/nodes/{first_node}",
+                          json=bad, timeout=10)
+        assert r.status_code == 200 and r.json()["applied"] == {"5": False}
+        from hivedscheduler_amd.sim import SimScheduler  # noqa: F401  (import check)
+        status = sched.get_physical_cluster_status()
+        def leaves(c):
+            kids = c.get("cellChildren") or []
+            return [c] if not kids else [l for k in kids for l in leaves(k)]
+        unhealthy = [c for c in leaves(status[0]) if c.get("cellHealthiness") != "Healthy"]
+        assert len(unhealthy) == 1, [c.get("cellAddress") for c in unhealthy]
+    finally:
+        server.should_exit = True
+        th.join(timeout=5)
