@@ -88,15 +88,22 @@ class CellProbeRunner:
         return os.path.exists(self.binary) and os.access(self.binary, os.X_OK)
 
     def probe_cell(self, leaf_indices: List[int], size_mb: int = 64, iters: int = 20,
-                   timeout_s: float = 120.0) -> dict:
+                   timeout_s: float = 300.0) -> dict:
         """All-reduce over the given GPU indices (one node). Returns the
-        parsed probe JSON plus a health verdict."""
+        parsed probe JSON plus a health verdict. A hung probe (RCCL init can
+        take minutes on a cold box; a truly wedged link hangs forever) is
+        killed at timeout_s and reported as not-ok rather than raising — the
+        agent loop must survive it."""
         env = dict(os.environ)
         env["HIP_VISIBLE_DEVICES"] = ",".join(str(i) for i in leaf_indices)
-        out = subprocess.run(
-            [self.binary, "--size-mb", str(size_mb), "--iters", str(iters)],
-            env=env, capture_output=True, text=True, timeout=timeout_s,
-        )
+        try:
+            out = subprocess.run(
+                [self.binary, "--size-mb", str(size_mb), "--iters", str(iters)],
+                env=env, capture_output=True, text=True, timeout=timeout_s,
+            )
+        except subprocess.TimeoutExpired:
+            return {"ok": False, "error": f"probe timed out after {timeout_s:.0f}s",
+                    "timeout": True, "leaf_indices": leaf_indices}
         if out.returncode != 0:
             return {"ok": False, "error": out.stderr.strip()[-2000:], "leaf_indices": leaf_indices}
         result = json.loads(out.stdout.strip().splitlines()[-1])
