@@ -52,3 +52,24 @@ def test_example_request_pods_parse_and_validate():
             assert spec.pinnedCellId
         if "opportunistic" in f:
             assert spec.priority == -1
+
+
+def test_example_basic_and_design_configs():
+    """The basic and heterogeneous design example configs parse into working
+    schedulers; the design config exercises pinned cells + multi-chain VCs."""
+    from hivedscheduler_amd.api import config as apicfg
+    from hivedscheduler_amd.sim import SimScheduler
+
+    basic = apicfg.init_raw_config(os.path.join(REPO, "examples/config/basic.yaml"))
+    sim = SimScheduler(basic)
+    assert sim.schedule("ns/b", sim.pod_spec(vc="default", leaf_cells=8)).kind == "bind"
+
+    design = apicfg.init_raw_config(
+        os.path.join(REPO, "examples/config/design-heterogeneous.yaml"))
+    sim2 = SimScheduler(design)
+    r = sim2.schedule("ns/pin", sim2.pod_spec(vc="VC1", leaf_cells=8,
+                                              pinned_cell_id="VC1-PIN"))
+    assert r.kind == "bind" and r.bind_info.node == "n4"
+    r2 = sim2.schedule("ns/ct1", sim2.pod_spec(vc="VC2", leaf_cells=2,
+                                               leaf_cell_type="CT1"))
+    assert r2.kind == "bind" and r2.bind_info.node in ("c1", "c2")
