@@ -7,6 +7,8 @@
 // an LCA-minimal, buddy-packed placement in O(depth * fanout) per pod — on the
 // fixed MI355X chain (leaf->pair->quad->node) this is a handful of comparisons,
 // which is what makes microsecond-scale Schedule() latency possible.
+#include <climits>
+
 #include "core.hpp"
 
 namespace hived {
@@ -67,6 +69,24 @@ std::pair<int, int> availLeaves(Cell* c, int p, PickSession& s) {
   return {at, af};
 }
 
+// Lowest LCA level achievable for q leaves anywhere inside `cell`
+// (INT_MAX if the subtree cannot hold q). This is the lookahead that makes
+// the descent LCA-minimal: a child with MORE availability may reach a LOWER
+// LCA (e.g. free = {0,1,2} in quad A vs {5,6} straddling quad B's pairs —
+// A serves a pair request at pair level, B only at quad level). Verified
+// against brute force in tests/test_placement_optimality.py.
+int bestLCALevel(Cell* c, int q, int p, PickSession& s) {
+  auto [t, f] = availLeaves(c, p, s);
+  (void)f;
+  if (t < q) return INT_MAX;
+  int best = c->level;
+  for (Cell* child : c->children) {
+    int b = bestLCALevel(child, q, p, s);
+    if (b < best) best = b;
+  }
+  return best;
+}
+
 // Pick q available leaf cells inside `cell`, minimizing the LCA level
 // (affinity) and, secondarily, the number of preemptions, with buddy-style
 // tight packing. Caller guarantees avail(cell) >= q.
@@ -80,20 +100,25 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
   std::vector<std::pair<int, int>> av(n);
   for (int i = 0; i < n; i++) av[i] = availLeaves(cell->children[i], p, s);
 
-  // A single child can hold the whole request: descend into the one that needs
-  // the fewest preemptions, then is the tightest fit (packing).
-  int best = -1;
+  // A single child can hold the whole request: descend into the one that
+  // reaches the lowest LCA, then needs the fewest preemptions, then is the
+  // tightest fit (packing).
+  int best = -1, bestLca = INT_MAX;
   for (int i = 0; i < n; i++) {
     if (av[i].first >= q) {
+      int lca = bestLCALevel(cell->children[i], q, p, s);
       if (best < 0) {
         best = i;
+        bestLca = lca;
         continue;
       }
       int needPreempt = std::max(0, q - av[i].second);
       int bestPreempt = std::max(0, q - av[best].second);
-      if (needPreempt < bestPreempt ||
-          (needPreempt == bestPreempt && av[i].first < av[best].first)) {
+      if (lca < bestLca ||
+          (lca == bestLca && (needPreempt < bestPreempt ||
+                              (needPreempt == bestPreempt && av[i].first < av[best].first)))) {
         best = i;
+        bestLca = lca;
       }
     }
   }

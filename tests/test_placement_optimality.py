@@ -1,0 +1,75 @@
+"""Property test: the C++ placement engine's best-fit descent claims to yield
+LCA-minimal placements (core/topo_sched.cpp). Verify against brute force:
+for random occupancy patterns on an 8-GPU node, the engine's placement must
+achieve the minimal possible LCA level among all choices of free leaves.
+
+LCA level of a placement = the level of the lowest common ancestor of the
+chosen leaves in the chain leaf(1) -> pair(2) -> quad(3) -> node(4); lower =
+tighter = better xGMI locality (reference topology_aware_scheduler.go:309-387
+achieves this by backtracking; ours by construction).
+"""
+import itertools
+import random
+
+from hivedscheduler_amd.sim import SimScheduler, mi355x_cluster_config
+
+
+def lca_level(indices):
+    """LCA level of leaf indices on the fixed 8-GPU chain."""
+    s = set(indices)
+    if len(s) == 1:
+        return 1
+    if any(s <= {2 * p, 2 * p + 1} for p in range(4)):
+        return 2
+    if s <= {0, 1, 2, 3} or s <= {4, 5, 6, 7}:
+        return 3
+    return 4
+
+
+def brute_force_best(free, q):
+    """Minimal achievable LCA level choosing q leaves from the free set."""
+    return min(lca_level(c) for c in itertools.combinations(sorted(free), q))
+
+
+def test_placement_is_lca_minimal_vs_brute_force():
+    rng = random.Random(42)
+    for trial in range(60):
+        sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+        # occupy a random subset of GPUs with 1-GPU pods
+        occupied = set(rng.sample(range(8), rng.randrange(0, 7)))
+        # occupy by scheduling singles then freeing the ones we don't want:
+        # simpler — schedule 8 singles, free the complement (placement of
+        # singles is deterministic: one leaf each)
+        placed = {}
+        for i in range(8):
+            r = sim.schedule(f"occ/{i}", sim.pod_spec(leaf_cells=1))
+            assert r.kind == "bind"
+            placed[r.bind_info.leafCellIsolation[0]] = f"occ/{i}"
+        for idx in set(range(8)) - occupied:
+            sim.delete_pod(placed[idx])
+        free = set(range(8)) - occupied
+        for q in (1, 2, 4):
+            if len(free) < q:
+                continue
+            r = sim.schedule(f"t/{trial}-{q}", sim.pod_spec(leaf_cells=q))
+            assert r.kind == "bind", (occupied, q)
+            got = r.bind_info.leafCellIsolation
+            assert set(got) <= free, (occupied, got)
+            assert lca_level(got) == brute_force_best(free, q), (
+                f"trial {trial}: occupied={sorted(occupied)} q={q} "
+                f"got={sorted(got)} (LCA {lca_level(got)}) vs optimal "
+                f"{brute_force_best(free, q)}")
+            sim.delete_pod(f"t/{trial}-{q}")
+
+
+def test_packing_preserves_future_big_requests():
+    """Buddy-style packing: after placing 1- and 2-GPU pods on a node, the
+    largest intact subtree must be as large as theoretically possible."""
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    assert sim.schedule("p/a", sim.pod_spec(leaf_cells=1)).kind == "bind"
+    assert sim.schedule("p/b", sim.pod_spec(leaf_cells=2)).kind == "bind"
+    assert sim.schedule("p/c", sim.pod_spec(leaf_cells=1)).kind == "bind"
+    # 4 GPUs used; a whole quad must still be free
+    r = sim.schedule("p/quad", sim.pod_spec(leaf_cells=4))
+    assert r.kind == "bind"
+    assert lca_level(r.bind_info.leafCellIsolation) == 3
