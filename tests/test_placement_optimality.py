@@ -73,3 +73,37 @@ def test_packing_preserves_future_big_requests():
     r = sim.schedule("p/quad", sim.pod_spec(leaf_cells=4))
     assert r.kind == "bind"
     assert lca_level(r.bind_info.leafCellIsolation) == 3
+
+
+def test_two_pass_prefers_free_over_preemption():
+    """A high-priority request that FITS in free cells never preempts, even
+    when preempting would give a tighter LCA (reference two-pass fit,
+    topology_aware_scheduler.go:82-92)."""
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    # low-priority pair on {0,1}; free: {2..7}
+    r0 = sim.schedule("lo/a", sim.pod_spec(priority=0, leaf_cells=2))
+    assert sorted(r0.bind_info.leafCellIsolation) == [0, 1]
+    # occupy 2,3 and 6,7 with another low-priority pods -> free {4,5}
+    assert sim.schedule("lo/b", sim.pod_spec(priority=0, leaf_cells=2)).kind == "bind"
+    assert sim.schedule("lo/c", sim.pod_spec(priority=0,
+                                             leaf_cells=2)).bind_info is not None
+    # high-priority pair: the free pair {4,5}? whichever pair remains free
+    r = sim.schedule("hi/x", sim.pod_spec(priority=10, leaf_cells=2))
+    assert r.kind == "bind"  # NOT preempt: free capacity suffices
+    free_pair = set(r.bind_info.leafCellIsolation)
+    assert lca_level(free_pair) == 2
+
+
+def test_preemption_pass_is_lca_minimal():
+    """When preemption IS needed, victims are chosen to keep the placement
+    LCA-tight (pair request preempts one pair, not halves of two pairs)."""
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    # fill the node with 4 low-priority pairs
+    for i in range(4):
+        assert sim.schedule(f"lo/{i}", sim.pod_spec(priority=0, leaf_cells=2)).kind == "bind"
+    r = sim.run_preemption_to_completion("hi/p", sim.pod_spec(priority=10, leaf_cells=2))
+    assert r.kind == "bind"
+    assert lca_level(r.bind_info.leafCellIsolation) == 2
+    # exactly one victim group died (gang semantics, minimal victims)
+    alive = [k for k in sim.pods if k.startswith("lo/")]
+    assert len(alive) == 3, alive
