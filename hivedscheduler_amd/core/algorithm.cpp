@@ -150,6 +150,7 @@ std::unique_ptr<Group> newGroup(const PodSpec& s, GState state) {
 
 ScheduleResult HivedCore::schedule(const PodSpec& s, const std::string& podKey,
                                    const std::set<std::string>& suggestedNodes, Phase phase) {
+  OpGuard opGuard(this);
   scheduleCount_++;
   Placement<PhysicalCell> phys;
   Placement<VirtualCell> virt;
@@ -524,6 +525,7 @@ bool HivedCore::scheduleOpportunisticGroup(const SchedulingRequest& sr,
 // ---------------------------------------------------------------------------
 
 void HivedCore::deleteUnallocatedPod(const PodSpec& s, const std::string& podKey) {
+  OpGuard opGuard(this);
   auto it = groups_.find(s.groupName);
   if (it == groups_.end() || it->second->state != GState::Preempting) return;
   Group* g = it->second.get();
@@ -534,6 +536,7 @@ void HivedCore::deleteUnallocatedPod(const PodSpec& s, const std::string& podKey
 }
 
 void HivedCore::addAllocatedPod(const PodSpec& s, const BindInfo& info, const std::string& podKey) {
+  OpGuard opGuard(this);
   int podIndex = 0;
   auto it = groups_.find(s.groupName);
   if (it != groups_.end()) {
@@ -554,6 +557,7 @@ void HivedCore::addAllocatedPod(const PodSpec& s, const BindInfo& info, const st
 
 void HivedCore::deleteAllocatedPod(const PodSpec& s, const BindInfo& info,
                                    const std::string& podKey) {
+  OpGuard opGuard(this);
   auto it = groups_.find(s.groupName);
   if (it == groups_.end()) return;
   Group* g = it->second.get();
@@ -609,6 +613,16 @@ void HivedCore::createAllocatedGroup(const PodSpec& s, const BindInfo& info,
           }
         } else {
           shouldLazyPreempt = shouldLazyPreempt || lazyPreempt;
+        }
+        if (p->reservingGroup != nullptr && p->reservingGroup != g) {
+          // cell event e8(i) (reference doc/design/state-machine.md): this
+          // cell is Reserved by a lower-priority Preempting group, but an
+          // Allocated group is now taking it (a higher-priority bind landed
+          // on the vacated reservation) -> cancel that preemptor entirely,
+          // releasing all its reservations; it will retry from Pending.
+          // Found by fuzzing: without this, the preemptor stays Preempting
+          // while its cells are Used by another group (state corruption).
+          deletePreemptingGroup(p->reservingGroup, podKey);
         }
         auto [safetyOk, reason] =
             allocateLeafCell(p, g->virtPlacement[leafCellNumber][podIndex][li], s.priority, g->vc);
@@ -864,6 +878,30 @@ std::pair<bool, std::string> HivedCore::allocateLeafCell(PhysicalCell* p, Virtua
     setCellPriority(p, priority);
     updateUsedLeafCellNumAtPriority(p, priority, true);
     VirtualCell* pac = v->preassigned;
+    if (pac->phys != nullptr && pac->phys->virt == pac &&
+        pac->phys->priority < kMinGuaranteedPriority &&
+        vcDoomedBadCells_[vc][p->chain].contains(pac->phys, pac->phys->level)) {
+      // The preassigned cell was doomed onto a bad physical cell AFTER this
+      // placement was computed against a different physical target (the doom
+      // machinery can fire mid-commit, e.g. inside an overlapping
+      // preemptor's cancellation). If the placement is not actually ON the
+      // doomed cell, undo the doomed binding so the real binding can form.
+      // Found by fuzzing: without this, the leaf binds under a preassigned
+      // pointing at an unrelated physical cell and the real target stays in
+      // the free list carrying a guaranteed priority.
+      PhysicalCell* anc = p;
+      while (anc->parent != nullptr && anc != pac->phys && anc->level < pac->phys->level) {
+        anc = static_cast<PhysicalCell*>(anc->parent);
+      }
+      if (anc != pac->phys) {
+        PhysicalCell* doomed = pac->phys;
+        pac->phys = nullptr;
+        doomed->virt = nullptr;
+        vcDoomedBadCells_[vc][p->chain].remove(doomed, doomed->level);
+        allVCDoomedBadCellNum_[p->chain][doomed->level]--;
+        releasePreassignedCell(doomed, vc, true);
+      }
+    }
     bool preassignedNewlyBound = pac->phys == nullptr;
     if (p->virt == nullptr) {
       // binding may already exist if the cell is bad

@@ -2,9 +2,18 @@
 // VC-safety accounting and bad-cell (doomed-bad) machinery.
 // Semantics parity: pkg/algorithm/cell_allocation.go:34-315 and
 // pkg/algorithm/hived_algorithm.go:453-653, 1354-1565.
+#include <cstdio>
+#include <cstdlib>
+
 #include "core.hpp"
 
 namespace hived {
+
+static bool mapDebug() {
+  static int v = -1;
+  if (v < 0) v = getenv("HIVED_DEBUG_MAP") ? 1 : 0;
+  return v == 1;
+}
 
 namespace {
 
@@ -12,7 +21,19 @@ struct AllocCtx {
   const std::set<std::string>& suggested;
   bool ignoreSuggested;
   std::unordered_map<VirtualCell*, PhysicalCell*>& bindings;
+  // physical cells (and their ancestors) claimed by earlier picks of THIS
+  // mapping round: lets the buddy-packing sort co-locate a group's cells
+  // (e.g. both pairs of a 2-pair group into ONE quad) so intact higher-level
+  // free cells — other VCs' guarantees — survive. Backtracking may leave
+  // stale entries; that only biases packing, never correctness.
+  std::unordered_set<PhysicalCell*>& claimed;
 };
+
+void markClaimed(const AllocCtx& ctx, PhysicalCell* c) {
+  for (PhysicalCell* a = c; a != nullptr; a = static_cast<PhysicalCell*>(a->parent)) {
+    ctx.claimed.insert(a);
+  }
+}
 
 // Usable = unbound, not a known-bad single-node cell, and (unless ignoring
 // suggestions) at least one node within the suggested set. Sorted by
@@ -37,7 +58,28 @@ std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& cand
     usable.push_back(c);
   }
   if (static_cast<int>(usable.size()) < numNeeded) return {};
-  std::stable_sort(usable.begin(), usable.end(), [](PhysicalCell* a, PhysicalCell* b) {
+  // Buddy packing first: prefer candidates in already-fragmented regions
+  // (fewer untouched free buddies), so intact higher-level free cells — other
+  // VCs' guarantees — survive. Fuzz-found: picking a pair under the last
+  // intact quad instead of one in a fragmented quad stranded a quad-level
+  // guarantee (totalLeft < allVCFree). Then opportunistic usage ascending,
+  // to minimize preemption of opportunistic pods (reference
+  // cell_allocation.go:252-315 ordering).
+  auto freeBuddies = [&ctx](PhysicalCell* c) {
+    if (c->parent == nullptr) return 0;
+    int n = 0;
+    for (Cell* b : c->parent->children) {
+      auto* pb = static_cast<PhysicalCell*>(b);
+      if (pb != c && pb->priority == kFreePriority && pb->virt == nullptr &&
+          !ctx.claimed.count(pb)) {
+        n++;
+      }
+    }
+    return n;
+  };
+  std::stable_sort(usable.begin(), usable.end(), [&](PhysicalCell* a, PhysicalCell* b) {
+    int fa = freeBuddies(a), fb = freeBuddies(b);
+    if (fa != fb) return fa < fb;
     return a->usedAt(kOpportunisticPriority) < b->usedAt(kOpportunisticPriority);
   });
   return usable;
@@ -65,6 +107,7 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cells,
       if (candidate->level == kLowestLevel) {
         ok = true;
         ctx.bindings[cells[cellIndex]->cell] = candidate;
+        markClaimed(ctx, candidate);
       } else {
         std::vector<BindingVertex*> childVerts;
         childVerts.reserve(cells[cellIndex]->children.size());
@@ -103,12 +146,21 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cells,
 // down. Backtracking handles cells that are bad or outside suggested nodes.
 bool buddyAlloc(BindingVertex* cell, ChainCellList& freeList, int currentLevel,
                 const AllocCtx& ctx) {
+  if (mapDebug()) {
+    fprintf(stderr, "[map] buddyAlloc virt level=%d at level=%d candidates:", cell->cell->level,
+            currentLevel);
+    for (Cell* c : freeList.at(currentLevel)) fprintf(stderr, " %s", c->address.c_str());
+    fprintf(stderr, "\n");
+  }
   if (currentLevel == cell->cell->level) {
     std::vector<PhysicalCell*> picked;
     if (mapVirtualCellsToPhysical({cell}, freeList.at(currentLevel), ctx, true, &picked)) {
       for (PhysicalCell* c : picked) freeList.remove(c, currentLevel);
+      if (mapDebug())
+        fprintf(stderr, "[map]  -> picked %s\n", picked.empty() ? "?" : picked[0]->address.c_str());
       return true;
     }
+    if (mapDebug()) fprintf(stderr, "[map]  -> no usable candidate at own level\n");
     return false;
   }
   std::vector<PhysicalCell*> freeCells = getUsablePhysicalCells(freeList.at(currentLevel), 1, ctx);
@@ -131,6 +183,7 @@ bool buddyAlloc(BindingVertex* cell, ChainCellList& freeList, int currentLevel,
 bool safeRelaxedBuddyAlloc(BindingVertex* cell, ChainCellList& freeList,
                            std::map<int, int>& freeCellNum, int currentLevel,
                            const AllocCtx& ctx) {
+  if (mapDebug()) fprintf(stderr, "[map] safeRelaxed for virt level=%d\n", currentLevel);
   int top = freeList.top();
   std::map<int, int> splittableNum;
   Cell* splittableCell = nullptr;
@@ -194,7 +247,8 @@ bool HivedCore::mapVirtualPlacementToPhysical(
     ChainCellList freeList, std::map<int, int> freeCellNum,
     const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
     std::unordered_map<VirtualCell*, PhysicalCell*>& bindings) {
-  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings};
+  std::unordered_set<PhysicalCell*> claimed;
+  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings, claimed};
   for (BindingVertex* c : preassigned) {
     if (!buddyAlloc(c, freeList, getLowestFreeCellLevel(freeList, c->cell->level), ctx)) {
       if (!safeRelaxedBuddyAlloc(c, freeList, freeCellNum, c->cell->level, ctx)) {
@@ -392,6 +446,7 @@ void HivedCore::releaseBadCell(PhysicalCell* c) {
 // --- node / cell health ------------------------------------------------------
 
 void HivedCore::setNodeHealthy(const std::string& node, bool healthy) {
+  OpGuard opGuard(this);
   if (healthy) {
     if (!badNodes_.count(node)) return;
     badNodes_.erase(node);
@@ -414,6 +469,7 @@ void HivedCore::setNodeHealthy(const std::string& node, bool healthy) {
 // any-child-bad roll-up. Node-level badness (badNodes_) takes precedence: a
 // leaf on a bad node cannot be marked healthy.
 void HivedCore::setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy) {
+  OpGuard opGuard(this);
   auto it = nodeLeafCellsStorage_.find(node);
   if (it == nodeLeafCellsStorage_.end()) return;
   for (PhysicalCell* leaf : it->second) {
@@ -494,10 +550,67 @@ void HivedCore::removeBadFreeCell(PhysicalCell* c) {
   tryUnbindDoomedBadCell(c->chain, c->level);
 }
 
+// Dooming allocates the bad physical cell out of the free list. badFreeCells_
+// contains SUBSUMED cells (free via an unsplit ancestor), and allocating one
+// of those splits every unsplit ancestor — removing a cell from totalLeft at
+// each split level — and removes the candidate's producible children below.
+// Found by fuzzing: dooming a bad pair under a free quad stranded another
+// VC's quad-level guarantee (totalLeft < allVCFree). A candidate is only
+// eligible if no level's guarantee would break.
+bool HivedCore::doomAllocationIsSafe(PhysicalCell* pc) {
+  const std::string& chain = pc->chain;
+  PhysicalCell* a = pc;
+  while (a->parent != nullptr && !static_cast<PhysicalCell*>(a->parent)->split) {
+    a = static_cast<PhysicalCell*>(a->parent);
+    if (totalLeftCellNum_[chain][a->level] - 1 < allVCFreeCellNum_[chain][a->level]) return false;
+  }
+  int numToReduce = static_cast<int>(pc->children.size());
+  for (int l = pc->level - 1; l >= kLowestLevel && numToReduce > 0; l--) {
+    if (totalLeftCellNum_[chain][l] - numToReduce < allVCFreeCellNum_[chain][l]) return false;
+    numToReduce *= static_cast<int>(fullCellList_[chain].at(l)[0]->children.size());
+  }
+  return true;
+}
+
+void HivedCore::tryBindDoomedBadCell(const std::string& chain, int level) {
+  if (inOperation_) {
+    pendingDoomChecks_.emplace_back(chain, level);
+    return;
+  }
+  doBindDoomedBadCell(chain, level);
+}
+
+void HivedCore::tryUnbindDoomedBadCell(const std::string& chain, int level) {
+  if (inOperation_) {
+    pendingDoomChecks_.emplace_back(chain, level);
+    return;
+  }
+  doUnbindDoomedBadCell(chain, level);
+}
+
+void HivedCore::drainDoomChecks() {
+  int guard = 0;
+  while (!pendingDoomChecks_.empty()) {
+    if (++guard > 100000) throw HivedError::Internal("doom-check drain did not converge");
+    auto [chain, level] = pendingDoomChecks_.front();
+    pendingDoomChecks_.erase(pendingDoomChecks_.begin());
+    // execute with the operation flag set so nested triggers re-queue
+    inOperation_ = true;
+    try {
+      doUnbindDoomedBadCell(chain, level);
+      doBindDoomedBadCell(chain, level);
+    } catch (...) {
+      inOperation_ = false;
+      throw;
+    }
+    inOperation_ = false;
+  }
+}
+
 // If healthy free cells < a VC's free cells at some level, some of the VC's
 // cells are inevitably bad: bind bad physical cells to free virtual cells so
 // users and the intra-VC scheduler can see them.
-void HivedCore::tryBindDoomedBadCell(const std::string& chain, int level) {
+void HivedCore::doBindDoomedBadCell(const std::string& chain, int level) {
   for (auto& [vcName, vcFreeNum] : vcFreeCellNum_) {
     auto chainIt = vcFreeNum.find(chain);
     if (chainIt == vcFreeNum.end()) continue;
@@ -505,7 +618,24 @@ void HivedCore::tryBindDoomedBadCell(const std::string& chain, int level) {
            totalLeftCellNum_[chain][level] -
                static_cast<int>(badFreeCells_[chain].at(level).size())) {
       if (badFreeCells_[chain].at(level).empty()) return;
-      auto* pc = static_cast<PhysicalCell*>(badFreeCells_[chain].at(level)[0]);
+      PhysicalCell* pc = nullptr;
+      for (Cell* cand : badFreeCells_[chain].at(level)) {
+        // Only cells DIRECTLY in the free list at this level are eligible:
+        // a subsumed candidate (free via an unsplit ancestor) would require
+        // splitting that ancestor, and tryBindDoomedBadCell can fire
+        // re-entrantly inside another allocation's free-list surgery — the
+        // split would mutate levels the outer operation is mid-way through
+        // (fuzz-found crash: "cell not found in list when removing").
+        // The ancestor itself is bad and listed at ITS level, so node-level
+        // dooming still covers the VC's higher-level quota.
+        if (!freeCellList_[chain].at(level).empty() &&
+            freeCellList_[chain].contains(cand, level) &&
+            doomAllocationIsSafe(static_cast<PhysicalCell*>(cand))) {
+          pc = static_cast<PhysicalCell*>(cand);
+          break;
+        }
+      }
+      if (pc == nullptr) return;  // no directly-free, safety-preserving candidate
       auto& pre = vcSchedulers_[vcName].nonPinnedPreassigned;
       auto preIt = pre.find(chain);
       VirtualCell* vc = nullptr;
@@ -522,7 +652,7 @@ void HivedCore::tryBindDoomedBadCell(const std::string& chain, int level) {
   }
 }
 
-void HivedCore::tryUnbindDoomedBadCell(const std::string& chain, int level) {
+void HivedCore::doUnbindDoomedBadCell(const std::string& chain, int level) {
   for (auto& [vcName, vcFreeNum] : vcFreeCellNum_) {
     auto chainIt = vcFreeNum.find(chain);
     if (chainIt == vcFreeNum.end()) continue;

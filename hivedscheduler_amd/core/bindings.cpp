@@ -418,6 +418,52 @@ class PyHivedCore {
   long long scheduleCount() const { return core_.scheduleCount_; }
   void checkInvariants() const { hived::checkInvariants(core_); }
 
+  py::dict debugCounters() const {
+    // safety-accounting introspection for tests/debugging: per chain+level
+    // totalLeft / allVCFree / per-VC free / bad-free count
+    py::dict d;
+    for (auto& [chain, perLevel] : core_.totalLeftCellNum_) {
+      py::dict cd;
+      for (auto& [level, left] : perLevel) {
+        py::dict ld;
+        ld["totalLeft"] = left;
+        int avf = 0;
+        auto ai = core_.allVCFreeCellNum_.find(chain);
+        if (ai != core_.allVCFreeCellNum_.end()) {
+          auto li = ai->second.find(level);
+          if (li != ai->second.end()) avf = li->second;
+        }
+        ld["allVCFree"] = avf;
+        py::dict vd;
+        for (auto& [vcn, perChain] : core_.vcFreeCellNum_) {
+          auto ci = perChain.find(chain);
+          if (ci == perChain.end()) continue;
+          auto li = ci->second.find(level);
+          if (li != ci->second.end() && li->second != 0) vd[py::str(vcn)] = li->second;
+        }
+        ld["vcFree"] = vd;
+        auto fi = core_.freeCellList_.find(chain);
+        if (fi != core_.freeCellList_.end() && level <= fi->second.top()) {
+          py::list fl;
+          for (auto* fc : fi->second.at(level)) fl.append(fc->address);
+          ld["freeList"] = fl;
+        }
+        auto di = core_.allVCDoomedBadCellNum_.find(chain);
+        if (di != core_.allVCDoomedBadCellNum_.end()) {
+          auto li2 = di->second.find(level);
+          if (li2 != di->second.end() && li2->second != 0) ld["doomed"] = li2->second;
+        }
+        auto bi = core_.badFreeCells_.find(chain);
+        if (bi != core_.badFreeCells_.end() && level <= bi->second.top()) {
+          ld["badFree"] = static_cast<int>(bi->second.at(level).size());
+        }
+        cd[py::int_(level)] = ld;
+      }
+      d[py::str(chain)] = cd;
+    }
+    return d;
+  }
+
  private:
   HivedCore core_;
 };
@@ -461,5 +507,6 @@ PYBIND11_MODULE(hivedcore, m) {
       .def("get_all_virtual_clusters_status", &PyHivedCore::getAllVirtualClustersStatus)
       .def("get_virtual_cluster_status", &PyHivedCore::getVirtualClusterStatus, py::arg("vc"))
       .def("schedule_count", &PyHivedCore::scheduleCount)
-      .def("check_invariants", &PyHivedCore::checkInvariants);
+      .def("check_invariants", &PyHivedCore::checkInvariants)
+      .def("debug_counters", &PyHivedCore::debugCounters);
 }

@@ -350,8 +350,39 @@ class HivedCore {
   void setHealthyCell(PhysicalCell* c);
   void addBadFreeCell(PhysicalCell* c);
   void removeBadFreeCell(PhysicalCell* c);
+  bool doomAllocationIsSafe(PhysicalCell* pc);
+  // Doomed-bad checks are DEFERRED to the end of the enclosing top-level
+  // operation (schedule commit, pod delete, health event): executing them
+  // mid-operation re-enters free-list surgery against a half-updated list
+  // and can consume cells an in-flight placement depends on (fuzz-found
+  // corruption). tryBind/tryUnbind queue when an operation is active; the
+  // bodies run from drainDoomChecks() once state is consistent.
   void tryBindDoomedBadCell(const std::string& chain, int level);
   void tryUnbindDoomedBadCell(const std::string& chain, int level);
+  void doBindDoomedBadCell(const std::string& chain, int level);
+  void doUnbindDoomedBadCell(const std::string& chain, int level);
+  void drainDoomChecks();
+
+  class OpGuard {
+   public:
+    explicit OpGuard(HivedCore* c)
+        : c_(c), outer_(!c->inOperation_), exceptionsAtEntry_(std::uncaught_exceptions()) {
+      c_->inOperation_ = true;
+    }
+    // On normal exit (including early returns) the outermost guard drains
+    // the deferred doom checks; on exception unwind it only resets the flag
+    // (queued checks stay pending and run at the next operation's end).
+    ~OpGuard() noexcept(false) {
+      if (!outer_) return;
+      c_->inOperation_ = false;
+      if (std::uncaught_exceptions() == exceptionsAtEntry_) c_->drainDoomChecks();
+    }
+
+   private:
+    HivedCore* c_;
+    bool outer_;
+    int exceptionsAtEntry_;
+  };
 
   // --- scheduling internals ---
   ScheduleResult generateResult(const Placement<PhysicalCell>& phys, bool hasVirtual,
@@ -429,6 +460,8 @@ class HivedCore {
   // state (public for inspect/bindings simplicity; external mutation forbidden)
   std::map<std::string, ChainCellList> fullCellList_;
   std::map<std::string, ChainCellList> freeCellList_;
+  bool inOperation_ = false;
+  std::vector<std::pair<std::string, int>> pendingDoomChecks_;
   std::map<std::string, IntraVCScheduler> vcSchedulers_;
   std::map<std::string, TopoScheduler> opportunisticSchedulers_;
   std::map<std::string, std::unique_ptr<Group>> groups_;
