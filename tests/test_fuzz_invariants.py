@@ -91,3 +91,86 @@ def test_fuzz_operations(seed):
                                                                leaf_cells=cells))
             assert r.kind == "bind", f"quota not restored for {vc} ({cells} cells): {r}"
     check()
+
+
+VCS_VARIANTS = [
+    VCS,
+    {"A": [("MI355X-NODE", 2)], "B": [("MI355X-NODE.MI355X-QUAD", 2)]},
+    {"A": [("MI355X-NODE.MI355X-QUAD.MI355X-PAIR", 3)],
+     "B": [("MI355X-NODE", 1)], "C": [("MI355X-NODE.MI355X-QUAD", 1)]},
+]
+
+
+@pytest.mark.parametrize("seed", list(range(8)))
+def test_fuzz_extended(seed):
+    """Wider fuzz: gang groups, priorities up to 10, varied cluster shapes,
+    preemption-protocol advancement/cancellation, health flapping — with the
+    full invariant check after every operation. This configuration found five
+    distinct core bugs during round 1 (see core/alloc.cpp, core/algorithm.cpp
+    comments marked "Found by fuzzing")."""
+    from hivedscheduler_amd.api.types import WebServerError
+
+    rng = random.Random(1000 + seed)
+    nnodes = rng.choice([3, 4])
+    vcs = rng.choice(VCS_VARIANTS)
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=nnodes, vcs=vcs))
+    check = sim.alg._core.check_invariants
+    live, preempting, counter = {}, {}, 0
+    nodes = sim.alg.all_nodes()
+    for step in range(400):
+        op = rng.random()
+        try:
+            if op < 0.38:
+                counter += 1
+                key = f"f/p{counter}"
+                kw = dict(vc=rng.choice(list(vcs)), priority=rng.choice([-1, -1, 0, 1, 2, 10]),
+                          leaf_cells=rng.choice([1, 2, 2, 4, 8]),
+                          lazy_preemption=rng.random() < 0.3)
+                if rng.random() < 0.25:
+                    kw["group"] = key
+                    kw["members"] = [(rng.choice([1, 2]), kw["leaf_cells"])]
+                spec = sim.pod_spec(**kw)
+                phase = "Filtering" if rng.random() < 0.6 else PREEMPTING
+                r = sim.schedule(key, spec, phase=phase)
+                if r.kind == "bind":
+                    live[key] = spec
+                elif r.kind == "preempt" and phase == PREEMPTING:
+                    preempting[key] = spec
+            elif op < 0.58 and live:
+                key = rng.choice(list(live))
+                del live[key]
+                sim.delete_pod(key)
+            elif op < 0.70 and preempting:
+                key = rng.choice(list(preempting))
+                spec = preempting[key]
+                if rng.random() < 0.3:
+                    sim.delete_unallocated(key, spec)
+                    del preempting[key]
+                else:
+                    r = sim.schedule(key, spec, phase=PREEMPTING)
+                    if r.kind == "preempt":
+                        for v in r.victim_pod_keys:
+                            if v in live:
+                                del live[v]
+                                sim.delete_pod(v)
+                    elif r.kind == "bind":
+                        live[key] = spec
+                        del preempting[key]
+                    else:
+                        del preempting[key]
+            elif op < 0.85:
+                sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
+            else:
+                sim.alg.set_leaf_cell_healthy(rng.choice(nodes), rng.randrange(8),
+                                              rng.random() < 0.7)
+            check()
+        except WebServerError:
+            continue  # 4xx user error (e.g. quota exceeded variants) is fine
+    # drain: all healthy, delete everything, tree must be invariant-clean
+    for node in nodes:
+        sim.alg.set_healthy_node(node)
+        for i in range(8):
+            sim.alg.set_leaf_cell_healthy(node, i, True)
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    check()
