@@ -3,9 +3,18 @@
 // replay (addAllocatedPod doubles as crash recovery and reconfiguration).
 // Semantics parity: pkg/algorithm/hived_algorithm.go:180-1352 and
 // pkg/algorithm/utils.go:38-310; state machines per doc/design/state-machine.md.
+#include <cstdio>
+#include <cstdlib>
+
 #include "core.hpp"
 
 namespace hived {
+
+bool mapDebugRelease() {
+  static int v = -1;
+  if (v < 0) v = getenv("HIVED_DEBUG_MAP") ? 1 : 0;
+  return v == 1;
+}
 
 namespace {
 
@@ -903,11 +912,37 @@ std::pair<bool, std::string> HivedCore::allocateLeafCell(PhysicalCell* p, Virtua
       }
     }
     bool preassignedNewlyBound = pac->phys == nullptr;
+    bool preassignedProducible = false;
+    if (preassignedNewlyBound) {
+      // The cell this bind will attach the preassigned to is p's ancestor at
+      // the preassigned's level. Only run the accounting allocation if that
+      // cell is actually producible from the free list: an earlier unbind
+      // whose matching release was skipped (priority roll-up from sibling
+      // subtrees kept the accounting "allocated") leaves the cell
+      // allocated-but-unbound — re-allocating it here would double-count
+      // and throw mid-commit (fuzz-found). Re-binding without re-allocating
+      // heals that state.
+      PhysicalCell* anc = p;
+      while (anc->parent != nullptr && anc->level < pac->level) {
+        anc = static_cast<PhysicalCell*>(anc->parent);
+      }
+      // authoritative membership: anc, or an unbound ancestor of it, must be
+      // DIRECTLY in the free list (inFreeCellList's split-flag shortcut
+      // reports stale true for an allocated-but-unbound cell)
+      auto& fl = freeCellList_[p->chain];
+      for (PhysicalCell* a = anc; a != nullptr && a->virt == nullptr;
+           a = static_cast<PhysicalCell*>(a->parent)) {
+        if (a->level <= fl.top() && fl.contains(a, a->level)) {
+          preassignedProducible = true;
+          break;
+        }
+      }
+    }
     if (p->virt == nullptr) {
       // binding may already exist if the cell is bad
       bindCell(p, v);
     }
-    if (preassignedNewlyBound) {
+    if (preassignedNewlyBound && preassignedProducible) {
       std::tie(safetyOk, reason) = allocatePreassignedCell(pac->phys, vc, false);
     }
   } else {
@@ -941,6 +976,15 @@ void HivedCore::releaseLeafCell(PhysicalCell* p, const std::string& vc) {
         !vcDoomedBadCells_[owner][preassignedPhysical->chain].contains(
             preassignedPhysical, preassignedPhysical->level)) {
       releasePreassignedCell(preassignedPhysical, owner, false);
+    } else if (mapDebugRelease()) {
+      fprintf(stderr, "[rel] keep preassigned %s: phys=%d pinned=%d prio=%d doomed=%d (leaf %s)\n",
+              v->preassigned->address.c_str(), preassignedPhysical != nullptr,
+              preassignedPhysical != nullptr && preassignedPhysical->pinned,
+              v->preassigned->priority,
+              preassignedPhysical != nullptr &&
+                  vcDoomedBadCells_[owner][preassignedPhysical->chain].contains(
+                      preassignedPhysical, preassignedPhysical->level),
+              p->address.c_str());
     }
   }
   updateUsedLeafCellNumAtPriority(p, p->priority, false);

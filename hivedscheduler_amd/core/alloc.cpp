@@ -636,6 +636,9 @@ void HivedCore::doBindDoomedBadCell(const std::string& chain, int level) {
         }
       }
       if (pc == nullptr) return;  // no directly-free, safety-preserving candidate
+      if (mapDebug())
+        fprintf(stderr, "[doom] bind %s level=%d for vc=%s\n", pc->address.c_str(), level,
+                vcName.c_str());
       auto& pre = vcSchedulers_[vcName].nonPinnedPreassigned;
       auto preIt = pre.find(chain);
       VirtualCell* vc = nullptr;
@@ -671,6 +674,9 @@ void HivedCore::doUnbindDoomedBadCell(const std::string& chain, int level) {
         }
       }
       if (pc == nullptr) break;
+      if (mapDebug())
+        fprintf(stderr, "[doom] unbind %s level=%d vc=%s\n", pc->address.c_str(), level,
+                vcName.c_str());
       pc->virt->phys = nullptr;
       pc->virt = nullptr;
       doomed.remove(pc, level);

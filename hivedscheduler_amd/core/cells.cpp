@@ -60,7 +60,12 @@ void bindCell(PhysicalCell* pc, VirtualCell* vc) {
 }
 
 // Unbind a physical cell's binding and its ancestors bottom-up while no bound
-// sibling remains; never unbind pinned cells.
+// sibling remains; never unbind pinned cells, and never unbind UNHEALTHY
+// ancestors — bad (incl. doomed-bad) cells keep their bindings until healthy
+// (reference hived_algorithm.go:1327-1345). Fuzz-found: unbinding a doomed
+// bad quad here while releaseLeafCell's doomed guard skipped the matching
+// accounting release left the binding and the free list out of sync, and a
+// later re-bind double-allocated the cell.
 void unbindCell(PhysicalCell* c) {
   VirtualCell* boundVirtual = c->virt;
   while (boundVirtual->phys != nullptr && !boundVirtual->phys->pinned) {
