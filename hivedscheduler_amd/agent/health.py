@@ -70,7 +70,9 @@ def collect_node_health(
 
         for i in range(n):
             try:
-                rep = gpu_health_report(i, quick=True, deep=sweep)
+                # best-of-3 bandwidth: a tenant job on the GPU can halve a
+                # single triad sample (see profiles/interference_r01.md)
+                rep = gpu_health_report(i, quick=True, deep=sweep, bw_samples=3)
                 report["gpus"][str(i)].update(
                     hbm_gbps=round(rep["hbm_gbps"], 1),
                     mfma_ok=rep["mfma_ok"],

@@ -21,13 +21,20 @@ def get_ops():
     return _ops
 
 
-def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False) -> dict:
+def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False,
+                      bw_samples: int = 1) -> dict:
     """Run the health-probe kernels on one GPU and return measured facts.
 
     Feeds leaf-cell healthiness: HBM bandwidth deficit, MFMA mismatch, or
     (deep=True) any stuck-bit error in the HBM pattern sweep marks the GPU's
     leaf cell bad. The deep sweep scans a bounded span (16 GiB quick / 64 GiB
     otherwise) of the 288 GB HBM3E in 4 GiB chunks.
+
+    bw_samples: take the BEST of N triad samples. On a GPU busy with a tenant
+    job a single sample can read under half the real bandwidth (measured:
+    1.53 TB/s vs 4.6 TB/s idle on the same healthy GPU — profiles/
+    interference_r01.md); genuinely sick HBM is low in EVERY sample. Agents
+    probing live nodes should use bw_samples>=3.
     """
     import torch
 
@@ -36,7 +43,10 @@ def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False) -
     size_mb = 256 if quick else 2048
     iters = 5 if quick else 20
     report = {"device": device, "info": ops.device_info(device)}
-    report["hbm_gbps"] = ops.hbm_triad_gbps(size_mb, iters)
+    samples = [ops.hbm_triad_gbps(size_mb, iters) for _ in range(max(1, bw_samples))]
+    report["hbm_gbps"] = max(samples)
+    if len(samples) > 1:
+        report["hbm_gbps_samples"] = [round(s, 1) for s in samples]
 
     torch.manual_seed(0)
     A = (torch.randn(16, 32, dtype=torch.float32) / 8).bfloat16().cuda(device)
