@@ -77,11 +77,29 @@ std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& cand
     }
     return n;
   };
+  // Same-parent-as-claimed is the strongest packing signal: when every
+  // candidate looks equally fragmented (e.g. all buddies squatted by
+  // opportunistic pods at priority -1, so freeBuddies ties at 0), the
+  // group's Nth cell must still land next to its earlier picks or the
+  // allocation splits one higher-level free cell per pod (fuzz-found
+  // VC-safety break).
+  auto parentClaimed = [&ctx](PhysicalCell* c) {
+    return c->parent != nullptr && ctx.claimed.count(static_cast<PhysicalCell*>(c->parent)) > 0;
+  };
   std::stable_sort(usable.begin(), usable.end(), [&](PhysicalCell* a, PhysicalCell* b) {
+    bool pa = parentClaimed(a), pb = parentClaimed(b);
+    if (pa != pb) return pa;
     int fa = freeBuddies(a), fb = freeBuddies(b);
     if (fa != fb) return fa < fb;
     return a->usedAt(kOpportunisticPriority) < b->usedAt(kOpportunisticPriority);
   });
+  if (mapDebug()) {
+    fprintf(stderr, "[sort]");
+    for (auto* u : usable)
+      fprintf(stderr, " %s(fb=%d,ot=%d)", u->address.c_str(), freeBuddies(u),
+              u->usedAt(kOpportunisticPriority));
+    fprintf(stderr, "\n");
+  }
   return usable;
 }
 
@@ -227,6 +245,9 @@ bool safeRelaxedBuddyAlloc(BindingVertex* cell, ChainCellList& freeList,
     std::vector<PhysicalCell*> picked;
     if (mapVirtualCellsToPhysical({cell}, cur, ctx, true, &picked)) {
       for (PhysicalCell* c : picked) freeList.remove(c, currentLevel);
+      if (mapDebug())
+        fprintf(stderr, "[map] safeRelaxed picked %s\n",
+                picked.empty() ? "?" : picked[0]->address.c_str());
       return true;
     }
   }
@@ -263,6 +284,35 @@ bool HivedCore::mapVirtualPlacementToPhysical(
     if (parentVirtual == nullptr || parentVirtual->phys == nullptr) return false;
     if (!mapVirtualCellsToPhysical(cells, parentVirtual->phys->children, ctx, false, nullptr)) {
       return false;
+    }
+  }
+  // HARD post-mapping safety guard: the mutated free-list COPY reflects the
+  // post-commit physical free state for this placement. At every level, the
+  // cells still producible from it must cover every OTHER VC's remaining free
+  // quota (allVCFree minus what this placement's new preassigned bindings
+  // consume). The packing heuristics above make placements good; this guard
+  // makes them SAFE — fuzzing kept finding candidate orders (bad cells +
+  // opportunistic squatters + relaxed splits) where a group's cells landed
+  // across multiple higher-level free cells and stranded another VC's
+  // guarantee. Rejecting here turns that into a correct "wait".
+  if (!preassigned.empty()) {
+    const std::string& chain = preassigned[0]->cell->chain;
+    auto avfIt = allVCFreeCellNum_.find(chain);
+    if (avfIt != allVCFreeCellNum_.end()) {
+      std::map<int, int> newPre;
+      for (BindingVertex* c : preassigned) newPre[c->cell->level]++;
+      int top = freeList.top();
+      int producible = 0;
+      for (int l = top; l >= kLowestLevel; l--) {
+        if (l < top && !fullCellList_[chain].at(l + 1).empty()) {
+          producible *= static_cast<int>(fullCellList_[chain].at(l + 1)[0]->children.size());
+        }
+        producible += static_cast<int>(freeList.at(l).size());
+        auto needIt = avfIt->second.find(l);
+        int need = needIt == avfIt->second.end() ? 0 : needIt->second;
+        need -= newPre.count(l) ? newPre[l] : 0;
+        if (producible < need) return false;
+      }
     }
   }
   return true;
@@ -514,10 +564,17 @@ void HivedCore::setHealthyCell(PhysicalCell* c) {
       c->virt = nullptr;
       vc->phys = nullptr;
       if (vc->parent == nullptr) {
-        // a preassigned cell: it must have been a doomed bad cell
-        vcDoomedBadCells_[vc->vc][c->chain].remove(c, c->level);
-        allVCDoomedBadCellNum_[c->chain][c->level]--;
-        releasePreassignedCell(c, vc->vc, true);
+        auto& doomed = vcDoomedBadCells_[vc->vc][c->chain];
+        if (c->level <= doomed.top() && doomed.contains(c, c->level)) {
+          // a doomed bad cell healing: undo the doom's accounting
+          doomed.remove(c, c->level);
+          allVCDoomedBadCellNum_[c->chain][c->level]--;
+          releasePreassignedCell(c, vc->vc, true);
+        }
+        // else: an allocated-bad preassigned whose pods already left — its
+        // accounting was released at pod-delete time while the binding was
+        // kept because the cell was bad (fuzz-found: assuming "doomed" here
+        // threw on the registry remove and double-released the accounting)
       }
     } else if (!c->pinned && vc->parent == nullptr &&
                vcDoomedBadCells_[vc->vc][c->chain].contains(c, c->level)) {
