@@ -48,9 +48,13 @@ void updateUsedLeafCellNumAtPriority(Cell* c, int p, bool increase) {
 }
 
 // Bind a virtual cell to a physical cell, and their ancestors bottom-up until
-// hitting an already-bound ancestor.
+// hitting an already-bound ancestor ON EITHER SIDE. The physical-side check
+// matters: walking past a physical ancestor that is already bound to a
+// DIFFERENT virtual cell would overwrite its pointer and strand the old
+// partner with a dangling phys reference (fuzz-found asymmetric binding
+// after a lazy-preemption revert whose victim kept bad-cell bindings).
 void bindCell(PhysicalCell* pc, VirtualCell* vc) {
-  while (vc->phys == nullptr) {
+  while (vc->phys == nullptr && pc->virt == nullptr) {
     pc->virt = vc;
     vc->phys = pc;
     if (vc->parent == nullptr) break;
