@@ -158,11 +158,21 @@ def test_fuzz_extended(seed):
                         del preempting[key]
                     else:
                         del preempting[key]
-            elif op < 0.85:
+            elif op < 0.83:
                 sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
-            else:
+            elif op < 0.97:
                 sim.alg.set_leaf_cell_healthy(rng.choice(nodes), rng.randrange(8),
                                               rng.random() < 0.7)
+            else:
+                # crash-recovery: fresh algorithm, replay live pods from bind
+                # infos (pods are the database; Preempting state is volatile)
+                new = SimScheduler(mi355x_cluster_config(num_nodes=nnodes, vcs=vcs))
+                for k, (sp, info) in list(sim.pods.items()):
+                    new.alg.add_allocated_pod(sp, info, k)
+                    new.pods[k] = (sp, info)
+                sim = new
+                check = sim.alg._core.check_invariants
+                preempting.clear()
             check()
         except WebServerError:
             continue  # 4xx user error (e.g. quota exceeded variants) is fine
