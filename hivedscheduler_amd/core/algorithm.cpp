@@ -848,6 +848,26 @@ std::pair<PhysicalCell*, VirtualCell*> HivedCore::findAllocatedLeafCell(
       if (t == preassignedType) preassignedLevel = l;
     }
   }
+  if (preassignedLevel > 0) {
+    // Ownership-conflict check (reconfiguration/recovery semantics,
+    // reference hived_algorithm.go:1036-1038): the recorded placement's
+    // preassigned-level cell may now sit INSIDE another preassigned's
+    // domain — e.g. after a restart, an earlier-replayed pod re-took a
+    // whole-node preassigned covering this pod's old quad (replay order is
+    // arbitrary). Binding across that boundary would double-own the
+    // subtree; fuzz-found via crash-recovery fuzzing. Lazy-preempt instead.
+    PhysicalCell* pre = p;
+    while (pre->parent != nullptr && pre->level < preassignedLevel) {
+      pre = static_cast<PhysicalCell*>(pre->parent);
+    }
+    for (PhysicalCell* a = static_cast<PhysicalCell*>(pre->parent); a != nullptr;
+         a = static_cast<PhysicalCell*>(a->parent)) {
+      if (a->virt != nullptr) {
+        *lazyPreempt = true;
+        return {p, nullptr};
+      }
+    }
+  }
   VirtualCell* v = nullptr;
   std::string message;
   if (preassignedLevel < 0) {
@@ -871,6 +891,13 @@ std::pair<PhysicalCell*, VirtualCell*> HivedCore::findAllocatedLeafCell(
     }
   }
   if (v == nullptr) {
+    *lazyPreempt = true;
+    return {p, nullptr};
+  }
+  if (v->vc != s.vc) {
+    // the physical cell is already bound into ANOTHER VC's virtual tree
+    // (cross-VC conflict after recovery/reconfiguration): never adopt a
+    // foreign binding — lazy-preempt this group instead
     *lazyPreempt = true;
     return {p, nullptr};
   }
