@@ -4,6 +4,7 @@
 // Semantics parity: pkg/algorithm/hived_algorithm.go:180-1352 and
 // pkg/algorithm/utils.go:38-310; state machines per doc/design/state-machine.md.
 #include <cstdio>
+#include <functional>
 #include <cstdlib>
 
 #include "core.hpp"
@@ -865,6 +866,26 @@ std::pair<PhysicalCell*, VirtualCell*> HivedCore::findAllocatedLeafCell(
       if (a->virt != nullptr) {
         *lazyPreempt = true;
         return {p, nullptr};
+      }
+    }
+    // inverse direction: an UNBOUND preassigned candidate with a bound
+    // descendant means another preassigned tree already owns part of this
+    // subtree (e.g. an earlier-replayed pod bound its own quad inside) —
+    // freshly binding `pre` on top would nest ownerships (fuzz-found:
+    // left the sibling quad in the free list while bound)
+    if (pre->virt == nullptr) {
+      std::function<bool(PhysicalCell*)> anyBound = [&](PhysicalCell* c) -> bool {
+        if (c->virt != nullptr) return true;
+        for (Cell* ch : c->children) {
+          if (anyBound(static_cast<PhysicalCell*>(ch))) return true;
+        }
+        return false;
+      };
+      for (Cell* ch : pre->children) {
+        if (anyBound(static_cast<PhysicalCell*>(ch))) {
+          *lazyPreempt = true;
+          return {p, nullptr};
+        }
       }
     }
   }

@@ -219,12 +219,20 @@ __global__ void pattern_verify_kernel(const unsigned long long* __restrict__ p, 
   if (local) atomicAdd(errors, local);
 }
 
-// Sweep up to max_gib GiB of this device's HBM in chunk_gib chunks.
+// Sweep up to max_gib GiB of this device's HBM in chunk_gib chunks. All
+// chunks are HELD until the sweep ends: a free/alloc loop would get the same
+// physical block back each time and re-scan one region. Coverage is bounded
+// by free memory minus headroom; allocation failure ends the sweep
+// gracefully (bytes_tested reports actual coverage).
 // Returns {bytes_tested, errors, write_gbps, verify_gbps}.
 py::dict hbm_sweep(long max_gib, long chunk_gib, long seed) {
   TORCH_CHECK(max_gib > 0 && chunk_gib > 0 && chunk_gib <= max_gib);
   const long chunk_bytes = chunk_gib << 30;
   const long n = chunk_bytes / 8;
+  size_t freeB = 0, totalB = 0;
+  HIP_CHECK(hipMemGetInfo(&freeB, &totalB));
+  const long headroom = 8L << 30;
+  long budget = std::min(max_gib << 30, (long)freeB - headroom);
   unsigned long long* errs_d = nullptr;
   HIP_CHECK(hipMalloc(&errs_d, sizeof(unsigned long long)));
   HIP_CHECK(hipMemset(errs_d, 0, sizeof(unsigned long long)));
@@ -235,10 +243,11 @@ py::dict hbm_sweep(long max_gib, long chunk_gib, long seed) {
   double write_s = 0, verify_s = 0;
   long tested = 0;
   int blocks = 8192, threads = 256;  // >> 256 CUs, fills all 8 XCDs
-  for (long off = 0; off + chunk_bytes <= (max_gib << 30); off += chunk_bytes) {
+  std::vector<unsigned long long*> chunks;
+  for (long off = 0; off + chunk_bytes <= budget; off += chunk_bytes) {
     unsigned long long* p = nullptr;
-    // keep headroom: stop if the allocator can't give another chunk
     if (hipMalloc(&p, chunk_bytes) != hipSuccess) { (void)hipGetLastError(); break; }
+    chunks.push_back(p);
     unsigned long long chunk_seed = (unsigned long long)seed ^ (unsigned long long)off;
     float ms = 0;
     HIP_CHECK(hipEventRecord(start, stream));
@@ -255,9 +264,9 @@ py::dict hbm_sweep(long max_gib, long chunk_gib, long seed) {
     HIP_CHECK(hipEventSynchronize(stop));
     HIP_CHECK(hipEventElapsedTime(&ms, start, stop));
     verify_s += ms / 1e3;
-    HIP_CHECK(hipFree(p));
     tested += chunk_bytes;
   }
+  for (unsigned long long* p : chunks) HIP_CHECK(hipFree(p));
   unsigned long long errs = 0;
   HIP_CHECK(hipMemcpy(&errs, errs_d, sizeof(errs), hipMemcpyDeviceToHost));
   HIP_CHECK(hipFree(errs_d));
