@@ -129,9 +129,14 @@ def test_fuzz_extended(seed):
                 if rng.random() < 0.25:
                     kw["group"] = key
                     kw["members"] = [(rng.choice([1, 2]), kw["leaf_cells"])]
+                if rng.random() < 0.3:
+                    kw["ignore_suggested"] = False
                 spec = sim.pod_spec(**kw)
                 phase = "Filtering" if rng.random() < 0.6 else PREEMPTING
-                r = sim.schedule(key, spec, phase=phase)
+                suggested = None
+                if rng.random() < 0.3:
+                    suggested = rng.sample(nodes, rng.randrange(1, len(nodes) + 1))
+                r = sim.schedule(key, spec, phase=phase, suggested=suggested)
                 if r.kind == "bind":
                     live[key] = spec
                 elif r.kind == "preempt" and phase == PREEMPTING:
