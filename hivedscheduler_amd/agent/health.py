@@ -49,14 +49,16 @@ def collect_node_health(
     probe_pairs: bool = False,
     sweep: bool = False,
 ) -> dict:
-    # NOTE on sweep: the stuck-bit scan transiently holds up to 16 GiB of
-    # HBM; on a GPU running tenant jobs this can make the tenant's own
-    # allocations fail. Keep sweep on the low-frequency cadence and prefer
-    # idle GPUs (the scheduler knows which leaves are unallocated).
     """One health sweep over all visible GPUs. deep=True also runs the HIP
     kernels (HBM + MFMA); sweep=True adds the 16 GiB HBM stuck-bit pattern
     sweep (any error => leaf bad); probe_pairs=True RCCL-probes each xGMI
-    pair."""
+    pair.
+
+    The stuck-bit scan transiently holds up to 16 GiB of HBM; on a GPU
+    running tenant jobs this can make the tenant's own allocations fail.
+    Keep sweep on the low-frequency cadence and prefer idle GPUs (the
+    scheduler knows which leaves are unallocated).
+    """
     report: dict = {"node": socket.gethostname(), "time": time.time(), "gpus": {}}
     alive = _rocm_smi_gpu_state()
     try:
