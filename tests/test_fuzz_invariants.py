@@ -189,3 +189,66 @@ def test_fuzz_extended(seed):
     for key in list(sim.pods):
         sim.delete_pod(key)
     check()
+
+
+@pytest.mark.parametrize("seed", list(range(4)))
+def test_fuzz_heterogeneous(seed):
+    """Two-chain fuzz (MI355X + CT1 SKUs, typed and typeless requests):
+    exercises chain iteration in scheduleNewAffinityGroup under churn."""
+    from hivedscheduler_amd.api import config as apicfg
+    from hivedscheduler_amd.api.types import (CellTypeSpec, Config, PhysicalCellSpec,
+                                              PhysicalClusterSpec, VirtualCellSpec,
+                                              VirtualClusterSpec, WebServerError)
+    from hivedscheduler_amd.topo.mi355x import mi355x_cell_types, mi355x_node_cell
+
+    ct = mi355x_cell_types()
+    ct["CT1-NODE"] = CellTypeSpec(childCellType="CT1", childCellNumber=2, isNodeLevel=True)
+    cfg = Config(
+        physicalCluster=PhysicalClusterSpec(
+            cellTypes=ct,
+            physicalCells=[mi355x_node_cell("node1"), mi355x_node_cell("node2"),
+                           PhysicalCellSpec(cellType="CT1-NODE", cellAddress="ct1"),
+                           PhysicalCellSpec(cellType="CT1-NODE", cellAddress="ct2")]),
+        virtualClusters={
+            "X": VirtualClusterSpec(virtualCells=[
+                VirtualCellSpec(cellType="MI355X-NODE", cellNumber=1),
+                VirtualCellSpec(cellType="CT1-NODE", cellNumber=1)]),
+            "Y": VirtualClusterSpec(virtualCells=[
+                VirtualCellSpec(cellType="MI355X-NODE.MI355X-QUAD", cellNumber=2),
+                VirtualCellSpec(cellType="CT1-NODE", cellNumber=1)]),
+        })
+    apicfg.infer_physical_cluster(cfg.physicalCluster)
+    rng = random.Random(2000 + seed)
+    sim = SimScheduler(cfg)
+    check = sim.alg._core.check_invariants
+    live, counter = {}, 0
+    nodes = sim.alg.all_nodes()
+    for step in range(300):
+        op = rng.random()
+        try:
+            if op < 0.45:
+                counter += 1
+                key = f"h/p{counter}"
+                kw = dict(vc=rng.choice(["X", "Y"]), priority=rng.choice([-1, 0, 1, 10]),
+                          leaf_cells=rng.choice([1, 2, 2, 4]))
+                if rng.random() < 0.5:
+                    kw["leaf_cell_type"] = rng.choice(["MI355X", "CT1"])
+                    if kw["leaf_cell_type"] == "CT1":
+                        kw["leaf_cells"] = rng.choice([1, 2])
+                r = sim.schedule(key, sim.pod_spec(**kw))
+                if r.kind == "bind":
+                    live[key] = kw
+            elif op < 0.75 and live:
+                key = rng.choice(list(live))
+                del live[key]
+                sim.delete_pod(key)
+            else:
+                sim.alg.set_leaf_cell_healthy(rng.choice(nodes),
+                                              rng.randrange(2 if rng.random() < 0.5 else 8),
+                                              rng.random() < 0.7)
+            check()
+        except WebServerError:
+            continue
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    check()
