@@ -73,3 +73,17 @@ def test_example_basic_and_design_configs():
     r2 = sim2.schedule("ns/ct1", sim2.pod_spec(vc="VC2", leaf_cells=2,
                                                leaf_cell_type="CT1"))
     assert r2.kind == "bind" and r2.bind_info.node in ("c1", "c2")
+
+
+def test_scenario_cli():
+    """The scenario replay CLI runs the example preemption scenario with all
+    expectations met (exit 0)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "hivedscheduler_amd.sim",
+         os.path.join(REPO, "examples/scenario-preemption.yaml")],
+        capture_output=True, text=True, timeout=120, cwd=REPO)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "0 expectation failure(s)" in out.stdout
