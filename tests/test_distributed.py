@@ -14,6 +14,10 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 def run_torchrun(script_args, nproc=2, timeout=300):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    # force the gloo/CPU path even on a GPU box: these tests model world>1
+    # on one host, and 2 ranks sharing 1 visible GPU breaks RCCL init
+    env["CUDA_VISIBLE_DEVICES"] = ""
+    env["HIP_VISIBLE_DEVICES"] = ""
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={nproc}",
