@@ -252,3 +252,64 @@ def test_fuzz_heterogeneous(seed):
     for key in list(sim.pods):
         sim.delete_pod(key)
     check()
+
+
+@pytest.mark.parametrize("seed", list(range(4)))
+def test_fuzz_pinned_cells(seed):
+    """Pinned-cell fuzz: requests with and without pinnedCellId against a
+    config with a statically pinned node, under churn + health flapping."""
+    from hivedscheduler_amd.api import config as apicfg
+    from hivedscheduler_amd.api.types import (Config, PhysicalClusterSpec,
+                                              PinnedCellSpec, VirtualCellSpec,
+                                              VirtualClusterSpec, WebServerError)
+    from hivedscheduler_amd.topo.mi355x import mi355x_cell_types, mi355x_node_cell
+
+    n1 = mi355x_node_cell("node1")
+    n1.pinnedCellId = "PIN1"
+    cfg = Config(
+        physicalCluster=PhysicalClusterSpec(
+            cellTypes=mi355x_cell_types(),
+            physicalCells=[n1, mi355x_node_cell("node2"), mi355x_node_cell("node3")]),
+        virtualClusters={
+            "P": VirtualClusterSpec(
+                virtualCells=[VirtualCellSpec(cellType="MI355X-NODE", cellNumber=1)],
+                pinnedCells=[PinnedCellSpec(pinnedCellId="PIN1")]),
+            "Q": VirtualClusterSpec(
+                virtualCells=[VirtualCellSpec(cellType="MI355X-NODE", cellNumber=1)]),
+        })
+    apicfg.infer_physical_cluster(cfg.physicalCluster)
+    rng = random.Random(3000 + seed)
+    sim = SimScheduler(cfg)
+    check = sim.alg._core.check_invariants
+    live, counter = {}, 0
+    nodes = sim.alg.all_nodes()
+    for step in range(300):
+        op = rng.random()
+        try:
+            if op < 0.45:
+                counter += 1
+                key = f"pin/p{counter}"
+                kw = dict(vc=rng.choice(["P", "P", "Q"]), priority=rng.choice([-1, 0, 1, 10]),
+                          leaf_cells=rng.choice([1, 2, 4, 8]))
+                if kw["vc"] == "P" and rng.random() < 0.5:
+                    kw["pinned_cell_id"] = "PIN1"
+                r = sim.schedule(key, sim.pod_spec(**kw))
+                if r.kind == "bind":
+                    live[key] = kw
+                    if kw.get("pinned_cell_id"):
+                        assert r.bind_info.node == "node1", r.bind_info.node
+            elif op < 0.75 and live:
+                key = rng.choice(list(live))
+                del live[key]
+                sim.delete_pod(key)
+            elif op < 0.9:
+                sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
+            else:
+                sim.alg.set_leaf_cell_healthy(rng.choice(nodes), rng.randrange(8),
+                                              rng.random() < 0.7)
+            check()
+        except WebServerError:
+            continue
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    check()
