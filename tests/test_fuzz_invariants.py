@@ -313,3 +313,67 @@ def test_fuzz_pinned_cells(seed):
     for key in list(sim.pods):
         sim.delete_pod(key)
     check()
+
+
+@pytest.mark.parametrize("seed", list(range(4)))
+def test_fuzz_reconfiguration_mutation(seed):
+    """Work-preserving reconfiguration fuzz: periodically restart into a
+    MUTATED config (grown cluster / reshuffled quotas) and replay all live
+    pods; mismatches must lazy-preempt or reject cleanly, never corrupt."""
+    from hivedscheduler_amd.api.types import WebServerError
+
+    rng = random.Random(4000 + seed)
+    nnodes = 3
+    vcs = {"VC1": [("MI355X-NODE", 1), ("MI355X-NODE.MI355X-QUAD", 1)],
+           "VC2": [("MI355X-NODE", 1), ("MI355X-NODE.MI355X-QUAD.MI355X-PAIR", 2)]}
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=nnodes, vcs=vcs))
+    check = sim.alg._core.check_invariants
+    live, counter = {}, 0
+    nodes = sim.alg.all_nodes()
+    for step in range(300):
+        op = rng.random()
+        try:
+            if op < 0.4:
+                counter += 1
+                key = f"rc/p{counter}"
+                kw = dict(vc=rng.choice(list(vcs)), priority=rng.choice([-1, 0, 1, 10]),
+                          leaf_cells=rng.choice([1, 2, 4, 8]))
+                r = sim.schedule(key, sim.pod_spec(**kw))
+                if r.kind == "bind":
+                    live[key] = kw
+            elif op < 0.65 and live:
+                key = rng.choice(list(live))
+                del live[key]
+                sim.delete_pod(key)
+            elif op < 0.9:
+                sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
+            else:
+                # mutate the cluster: grow, or shift a quad of quota
+                mode = rng.randrange(3)
+                if mode == 0:
+                    nnodes += 1
+                elif mode == 1:
+                    vcs = {k: list(v) for k, v in vcs.items()}
+                    vcs["VC1"] = vcs["VC1"] + [("MI355X-NODE.MI355X-QUAD", 1)]
+                try:
+                    new = SimScheduler(mi355x_cluster_config(num_nodes=nnodes, vcs=vcs))
+                except WebServerError:
+                    continue  # quota no longer fits; keep the old cluster
+                for k, (sp, info) in list(sim.pods.items()):
+                    try:
+                        new.alg.add_allocated_pod(sp, info, k)
+                        new.pods[k] = (sp, info)
+                    except WebServerError:
+                        pass
+                sim = new
+                check = sim.alg._core.check_invariants
+                nodes = sim.alg.all_nodes()
+                for n in nodes:
+                    sim.alg.set_healthy_node(n)
+                live = {k: live[k] for k in live if k in sim.pods}
+            check()
+        except WebServerError:
+            continue
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    check()
