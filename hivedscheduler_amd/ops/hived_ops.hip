@@ -45,8 +45,8 @@ __global__ void triad_kernel(const float4* __restrict__ a, const float4* __restr
 }
 
 // Returns achieved GB/s (3 streams: 2 reads + 1 write).
-double hbm_triad_gbps(long size_mb, long iters) {
-  TORCH_CHECK(size_mb > 0 && iters > 0);
+double hbm_triad_gbps(long size_mb, long iters, long blocks, long threads) {
+  TORCH_CHECK(size_mb > 0 && iters > 0 && blocks > 0 && threads > 0 && threads <= 1024);
   long bytes = size_mb * 1024 * 1024;
   long n = bytes / sizeof(float4);
   auto opts = torch::TensorOptions().dtype(torch::kFloat32).device(torch::kCUDA);
@@ -54,8 +54,6 @@ double hbm_triad_gbps(long size_mb, long iters) {
   auto b = torch::ones({bytes / 4}, opts);
   auto c = torch::empty({bytes / 4}, opts);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
-  int blocks = 8192;  // >> 256 CUs to fill all 8 XCDs
-  int threads = 256;
   // warmup
   hipLaunchKernelGGL(triad_kernel, dim3(blocks), dim3(threads), 0, stream,
                      reinterpret_cast<const float4*>(a.data_ptr<float>()),
@@ -302,7 +300,10 @@ int device_count() {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X GPU health-probe kernels (gfx950 HIP)";
+  // default geometry from an on-hardware sweep (profiles/interference_r01.md):
+  // 32768x1024 sustains 5.6 TB/s vs 5.0-5.2 at 8192x256 (+10%)
   m.def("hbm_triad_gbps", &hbm_triad_gbps, py::arg("size_mb") = 1024, py::arg("iters") = 10,
+        py::arg("blocks") = 32768, py::arg("threads") = 1024,
         "Streaming HBM bandwidth in GB/s (triad: 2 reads + 1 write)");
   m.def("mfma_check", &mfma_check, py::arg("A"), py::arg("B"), py::arg("blocks") = 2048,
         py::arg("repeats") = 1, "Per-wave bf16 MFMA tile GEMM across all CUs");
