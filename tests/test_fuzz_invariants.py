@@ -377,3 +377,85 @@ def test_fuzz_reconfiguration_mutation(seed):
     for key in list(sim.pods):
         sim.delete_pod(key)
     check()
+
+
+@pytest.mark.parametrize("seed", list(range(4)))
+def test_fuzz_rack_chains(seed):
+    """Five-level chain fuzz (leaf->pair->quad->node->rack): exercises buddy
+    allocation and safety accounting at the pool level, plus crash-recovery
+    replay on rack configs."""
+    from hivedscheduler_amd.api import config as apicfg
+    from hivedscheduler_amd.api.types import (Config, PhysicalCellSpec,
+                                              PhysicalClusterSpec, VirtualCellSpec,
+                                              VirtualClusterSpec, WebServerError)
+    from hivedscheduler_amd.topo.mi355x import mi355x_cell_types, mi355x_node_cell
+
+    rng = random.Random(5000 + seed)
+    n_racks = rng.choice([2, 3])
+    ct = mi355x_cell_types(pool_sizes=(2,))
+    cells, idx = [], 0
+    for _ in range(n_racks):
+        kids = []
+        for _ in range(2):
+            idx += 1
+            kids.append(mi355x_node_cell(f"node{idx}"))
+        cells.append(PhysicalCellSpec(cellType="2-MI355X-NODE", cellChildren=kids))
+    cfg = Config(
+        physicalCluster=PhysicalClusterSpec(cellTypes=ct, physicalCells=cells),
+        virtualClusters={
+            "A": VirtualClusterSpec(virtualCells=[
+                VirtualCellSpec(cellType="2-MI355X-NODE", cellNumber=n_racks - 1)]),
+            "B": VirtualClusterSpec(virtualCells=[
+                VirtualCellSpec(cellType="2-MI355X-NODE.MI355X-NODE", cellNumber=1),
+                VirtualCellSpec(cellType="2-MI355X-NODE.MI355X-NODE.MI355X-QUAD", cellNumber=1)]),
+        })
+    apicfg.infer_physical_cluster(cfg.physicalCluster)
+    sim = SimScheduler(cfg)
+    check = sim.alg._core.check_invariants
+    live, counter = {}, 0
+    nodes = sim.alg.all_nodes()
+    for step in range(300):
+        op = rng.random()
+        try:
+            if op < 0.42:
+                counter += 1
+                key = f"rk/p{counter}"
+                kw = dict(vc=rng.choice(["A", "B"]), priority=rng.choice([-1, 0, 1, 10]),
+                          leaf_cells=rng.choice([1, 2, 4, 8]),
+                          group=None)
+                if rng.random() < 0.3:
+                    kw["group"] = key
+                    kw["members"] = [(2, kw["leaf_cells"])]
+                else:
+                    kw.pop("group")
+                r = sim.schedule(key, sim.pod_spec(**kw))
+                if r.kind == "bind":
+                    live[key] = kw
+            elif op < 0.7 and live:
+                key = rng.choice(list(live))
+                del live[key]
+                sim.delete_pod(key)
+            elif op < 0.9:
+                sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
+            elif op < 0.97:
+                sim.alg.set_leaf_cell_healthy(rng.choice(nodes), rng.randrange(8),
+                                              rng.random() < 0.7)
+            else:
+                new = SimScheduler(cfg)
+                for k, (sp, info) in list(sim.pods.items()):
+                    try:
+                        new.alg.add_allocated_pod(sp, info, k)
+                        new.pods[k] = (sp, info)
+                    except WebServerError:
+                        pass
+                sim = new
+                check = sim.alg._core.check_invariants
+                for n in nodes:
+                    sim.alg.set_healthy_node(n)
+                live = {k: live[k] for k in live if k in sim.pods}
+            check()
+        except WebServerError:
+            continue
+    for key in list(sim.pods):
+        sim.delete_pod(key)
+    check()
