@@ -196,10 +196,32 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
     return a.usedHigherPriority < b.usedHigherPriority;
   });
 
-  // Greedy gang fit of pods (ascending leaf-cell number) onto sorted nodes.
+  // Single-node preference: when ONE node can host the whole gang, use it
+  // (most-packed first via the sort above). The reference's greedy fit
+  // splits a gang as soon as the first packed node runs out (topology_
+  // aware_scheduler.go:268-306) — acceptable on NVSwitch, but on MI355X a
+  // split gang communicates over the NETWORK while a same-node gang rides
+  // 7x ~153 GB/s xGMI links, so locality outranks strict packing here.
+  // (Property-tested: tests/test_placement_optimality.py.)
   std::vector<int> pickedNodeIndices(sortedLeafNums.size(), -1);
   size_t podIndex = 0;
   int pickedLeafCellNum = 0;
+  int totalLeafNum = 0;
+  for (int q : sortedLeafNums) totalLeafNum += q;
+  for (size_t nodeIndex = 0; nodeIndex < cv.size(); nodeIndex++) {
+    const NodeView& n = cv[nodeIndex];
+    if (n.freeAtPriority >= totalLeafNum) {
+      if (!n.suggested) {
+        *failedReason = "have to use at least one non-suggested node " + n.c->address;
+        return false;
+      }
+      for (size_t i = 0; i < sortedLeafNums.size(); i++) {
+        pickedNodeIndices[i] = static_cast<int>(nodeIndex);
+      }
+      podIndex = sortedLeafNums.size();
+      break;
+    }
+  }
   for (size_t nodeIndex = 0; nodeIndex < cv.size() && podIndex < sortedLeafNums.size();) {
     const NodeView& n = cv[nodeIndex];
     if (n.freeAtPriority - pickedLeafCellNum >= sortedLeafNums[podIndex]) {

@@ -107,3 +107,40 @@ def test_preemption_pass_is_lca_minimal():
     # exactly one victim group died (gang semantics, minimal victims)
     alive = [k for k in sim.pods if k.startswith("lo/")]
     assert len(alive) == 3, alive
+
+
+def test_gang_never_spreads_when_one_node_suffices():
+    """Across random multi-node occupancies: if ANY single node can host the
+    whole gang, the gang lands on one node (greedy fit over sorted nodes,
+    reference topology_aware_scheduler.go:268-306)."""
+    rng = random.Random(99)
+    for trial in range(30):
+        sim = SimScheduler(mi355x_cluster_config(
+            num_nodes=3, vcs={"VC1": [("MI355X-NODE", 3)]}))
+        # random occupancy: singles sprinkled across nodes
+        placed = []
+        for i in range(rng.randrange(0, 12)):
+            r = sim.schedule(f"occ/{trial}-{i}", sim.pod_spec(leaf_cells=1))
+            if r.kind == "bind":
+                placed.append(f"occ/{trial}-{i}")
+        # free some randomly
+        for k in rng.sample(placed, rng.randrange(0, len(placed) + 1) if placed else 0):
+            sim.delete_pod(k)
+        # per-node free counts
+        free = {}
+        for top in sim.alg.get_physical_cluster_status():
+            node = top["cellAddress"]
+            def count_free(c):
+                kids = c.get("cellChildren") or []
+                if not kids:
+                    return 1 if c["cellState"] == "Free" else 0
+                return sum(count_free(k) for k in kids)
+            free[node] = count_free(top)
+        spec = sim.pod_spec(leaf_cells=2, group=f"g{trial}", members=[(2, 2)])
+        r1 = sim.schedule(f"g/{trial}-0", spec)
+        r2 = sim.schedule(f"g/{trial}-1", spec)
+        if max(free.values()) >= 4:
+            assert r1.kind == "bind" and r2.kind == "bind", (free, r1, r2)
+            assert r1.bind_info.node == r2.bind_info.node, (
+                f"trial {trial}: gang split {r1.bind_info.node}/{r2.bind_info.node} "
+                f"with free={free}")
