@@ -58,7 +58,16 @@ def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False,
     report["mfma_max_err_vs_fp32"] = max_err
     report["mfma_cross_cu_spread"] = tile_spread  # must be exactly 0
     report["mfma_ok"] = bool(tile_spread == 0.0 and max_err < 0.1)
-    report["healthy"] = bool(report["mfma_ok"] and report["hbm_gbps"] > 1000.0)
+    # CU coverage: a big grid must place waves on every CU of every XCD; a
+    # fused-off / hung CU shows up as missing (xcc, se, sh, cu) tuples
+    words = ops.cu_coverage(4096).cpu().numpy()
+    cus = {(int(w) >> 16, (int(w) >> 13) & 0x7, (int(w) >> 12) & 0x1, (int(w) >> 8) & 0xF)
+           for w in words}
+    report["cu_coverage"] = len(cus)
+    expected_cus = report["info"]["multiProcessorCount"]
+    report["cu_coverage_ok"] = bool(len(cus) >= expected_cus)
+    report["healthy"] = bool(report["mfma_ok"] and report["cu_coverage_ok"]
+                             and report["hbm_gbps"] > 1000.0)
     if deep:
         sweep = ops.hbm_sweep(16 if quick else 64, 4, 1)
         report["hbm_sweep"] = sweep

@@ -139,6 +139,34 @@ torch::Tensor mfma_check(torch::Tensor A, torch::Tensor B, long blocks, long rep
 }
 
 // ---------------------------------------------------------------------------
+// CU coverage: record each wave's hardware placement so the host can verify
+// that a health-check grid actually touched every CU on every XCD (a hung or
+// fused-off CU shows up as missing coverage). HW_ID (gfx9-family layout):
+// CU_ID[11:8], SH_ID[12], SE_ID[15:13]; XCC_ID identifies the XCD (0-7).
+// ---------------------------------------------------------------------------
+__global__ void cu_coverage_kernel(unsigned* __restrict__ out) {
+  unsigned hwid, xcc;
+  asm volatile("s_getreg_b32 %0, hwreg(HW_REG_HW_ID)" : "=s"(hwid));
+  asm volatile("s_getreg_b32 %0, hwreg(HW_REG_XCC_ID)" : "=s"(xcc));
+  int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if ((threadIdx.x & 63) == 0) out[wave] = (xcc << 16) | (hwid & 0xffff);
+}
+
+// Returns an int32 tensor of per-wave (XCC_ID<<16 | HW_ID) words; the host
+// derives the set of distinct (xcc, se, sh, cu) tuples.
+torch::Tensor cu_coverage(long blocks) {
+  TORCH_CHECK(blocks > 0);
+  int threads = 256;
+  long waves = blocks * (threads / 64);
+  auto out = torch::zeros({waves}, torch::TensorOptions().dtype(torch::kInt32).device(torch::kCUDA));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(cu_coverage_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     reinterpret_cast<unsigned*>(out.data_ptr<int>()));
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // xGMI p2p bandwidth between two devices (one direction).
 // ---------------------------------------------------------------------------
 double p2p_gbps(int src_dev, int dst_dev, long size_mb, long iters) {
@@ -312,6 +340,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hbm_sweep", &hbm_sweep, py::arg("max_gib") = 16, py::arg("chunk_gib") = 4,
         py::arg("seed") = 1,
         "Stuck-bit pattern sweep over up to max_gib GiB of HBM; returns error count");
+  m.def("cu_coverage", &cu_coverage, py::arg("blocks") = 4096,
+        "Per-wave (XCC_ID<<16 | HW_ID) placement words for CU-coverage checks");
   m.def("device_info", &device_info, py::arg("dev") = 0);
   m.def("device_count", &device_count);
 }

@@ -207,3 +207,19 @@ def test_agent_end_to_end_on_hardware():
     finally:
         server.should_exit = True
         th.join(timeout=5)
+
+
+@needs_gpu
+def test_cu_coverage_all_256():
+    """The coverage grid must place waves on every CU of every XCD: 256
+    distinct (xcc, se, sh, cu) tuples on MI355X (8 XCDs x 32 CUs)."""
+    from hivedscheduler_amd.ops import get_ops
+
+    ops = get_ops()
+    words = ops.cu_coverage(4096).cpu().numpy()
+    xccs = {int(w) >> 16 for w in words}
+    cus = {(int(w) >> 16, (int(w) >> 13) & 0x7, (int(w) >> 12) & 0x1, (int(w) >> 8) & 0xF)
+           for w in words}
+    n = get_ops().device_info(0)["multiProcessorCount"]
+    assert len(xccs) == 8, f"XCDs covered: {sorted(xccs)}"
+    assert len(cus) >= n, f"CUs covered: {len(cus)} < {n}"
