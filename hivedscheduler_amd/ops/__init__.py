@@ -66,7 +66,9 @@ def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False,
     report["cu_coverage"] = len(cus)
     expected_cus = report["info"]["multiProcessorCount"]
     report["cu_coverage_ok"] = bool(len(cus) >= expected_cus)
+    report["lds_errors"] = int(ops.lds_check(2048, 7))
     report["healthy"] = bool(report["mfma_ok"] and report["cu_coverage_ok"]
+                             and report["lds_errors"] == 0
                              and report["hbm_gbps"] > 1000.0)
     if deep:
         sweep = ops.hbm_sweep(16 if quick else 64, 4, 1)

@@ -139,6 +139,46 @@ torch::Tensor mfma_check(torch::Tensor A, torch::Tensor B, long blocks, long rep
 }
 
 // ---------------------------------------------------------------------------
+// LDS integrity: each workgroup fills a 64 KB LDS slab with address-derived
+// patterns, barriers, and read-verifies from different lanes (catches
+// per-CU LDS faults; HBM and matrix pipes are covered by the other checks).
+// ---------------------------------------------------------------------------
+__global__ void lds_check_kernel(unsigned long long* __restrict__ errors,
+                                 unsigned long long seed) {
+  __shared__ unsigned lds[16384];  // 64 KB
+  int t = threadIdx.x;
+  for (int i = t; i < 16384; i += blockDim.x) {
+    lds[i] = (unsigned)(seed ^ (blockIdx.x * 16384u + i) * 2654435761u);
+  }
+  __syncthreads();
+  // verify with a shifted lane mapping so each value is read by a different
+  // thread than wrote it (exercises cross-lane LDS paths)
+  unsigned long long local = 0;
+  for (int i = t; i < 16384; i += blockDim.x) {
+    int j = (i + 4097) & 16383;
+    unsigned want = (unsigned)(seed ^ (blockIdx.x * 16384u + j) * 2654435761u);
+    if (lds[j] != want) local++;
+  }
+  if (local) atomicAdd(errors, local);
+}
+
+// Run the LDS slab check across a grid >> 256 CUs; returns error count.
+long lds_check(long blocks, long seed) {
+  TORCH_CHECK(blocks > 0);
+  unsigned long long* errs_d = nullptr;
+  HIP_CHECK(hipMalloc(&errs_d, sizeof(unsigned long long)));
+  HIP_CHECK(hipMemset(errs_d, 0, sizeof(unsigned long long)));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(lds_check_kernel, dim3(blocks), dim3(256), 0, stream, errs_d,
+                     (unsigned long long)seed);
+  HIP_CHECK(hipGetLastError());
+  unsigned long long errs = 0;
+  HIP_CHECK(hipMemcpy(&errs, errs_d, sizeof(errs), hipMemcpyDeviceToHost));
+  HIP_CHECK(hipFree(errs_d));
+  return (long)errs;
+}
+
+// ---------------------------------------------------------------------------
 // CU coverage: record each wave's hardware placement so the host can verify
 // that a health-check grid actually touched every CU on every XCD (a hung or
 // fused-off CU shows up as missing coverage). HW_ID (gfx9-family layout):
@@ -340,6 +380,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hbm_sweep", &hbm_sweep, py::arg("max_gib") = 16, py::arg("chunk_gib") = 4,
         py::arg("seed") = 1,
         "Stuck-bit pattern sweep over up to max_gib GiB of HBM; returns error count");
+  m.def("lds_check", &lds_check, py::arg("blocks") = 2048, py::arg("seed") = 1,
+        "LDS slab write/read-verify across the chip; returns error count");
   m.def("cu_coverage", &cu_coverage, py::arg("blocks") = 4096,
         "Per-wave (XCC_ID<<16 | HW_ID) placement words for CU-coverage checks");
   m.def("device_info", &device_info, py::arg("dev") = 0);

@@ -223,3 +223,14 @@ def test_cu_coverage_all_256():
     n = get_ops().device_info(0)["multiProcessorCount"]
     assert len(xccs) == 8, f"XCDs covered: {sorted(xccs)}"
     assert len(cus) >= n, f"CUs covered: {len(cus)} < {n}"
+
+
+@needs_gpu
+def test_lds_check_clean():
+    """LDS slab write/read-verify across the chip: zero errors on healthy
+    hardware, and the full health report includes it."""
+    from hivedscheduler_amd.ops import get_ops, gpu_health_report
+
+    assert get_ops().lds_check(2048, 99) == 0
+    rep = gpu_health_report(0, quick=True)
+    assert rep["lds_errors"] == 0 and rep["healthy"]
