@@ -82,6 +82,8 @@ def collect_node_health(
                 report["gpus"][str(i)].update(
                     hbm_gbps=round(rep["hbm_gbps"], 1),
                     mfma_ok=rep["mfma_ok"],
+                    cu_coverage=rep["cu_coverage"],
+                    lds_errors=rep["lds_errors"],
                 )
                 if "hbm_sweep" in rep:
                     report["gpus"][str(i)]["hbm_sweep_errors"] = rep["hbm_sweep"]["errors"]
