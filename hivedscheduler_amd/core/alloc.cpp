@@ -543,12 +543,20 @@ void HivedCore::setBadCell(PhysicalCell* c) {
   if (c->parent != nullptr) setBadCell(static_cast<PhysicalCell*>(c->parent));
   if (inFreeCellList(c)) {
     addBadFreeCell(c);
-  } else if (c->virt == nullptr && !c->split) {
-    // an ancestor is bound to a virtual cell; bind this bad cell too so the
-    // VC scheduler can see the failure
-    VirtualCell* vc = getUnboundVirtualCell(static_cast<PhysicalCell*>(c->parent)->virt->children);
-    c->virt = vc;
-    vc->phys = c;
+  } else if (c->virt == nullptr && !c->split && c->parent != nullptr &&
+             static_cast<PhysicalCell*>(c->parent)->virt != nullptr) {
+    // the parent is bound to a virtual cell: bind this bad cell to one of
+    // its unbound virtual children too, so the VC scheduler can see the
+    // failure. The parent may be UNBOUND (a deeper ancestor holds the
+    // binding) or every virtual child slot may already be taken — in either
+    // case skip: the exposure binding is best-effort visibility, and
+    // dereferencing a missing slot segfaulted (fuzz seed 1431).
+    VirtualCell* vc =
+        getUnboundVirtualCell(static_cast<PhysicalCell*>(c->parent)->virt->children);
+    if (vc != nullptr) {
+      c->virt = vc;
+      vc->phys = c;
+    }
   }
 }
 
