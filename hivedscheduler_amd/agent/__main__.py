@@ -6,7 +6,9 @@ import logging
 
 def main() -> None:
     ap = argparse.ArgumentParser(prog="hivedscheduler-amd-agent")
-    ap.add_argument("--scheduler", required=True, help="scheduler base URL")
+    ap.add_argument("--scheduler", default=None, help="scheduler base URL")
+    ap.add_argument("--local", action="store_true",
+                    help="one local health sweep (deep + sweep), print JSON, no posting")
     ap.add_argument("--node-name", default=None)
     ap.add_argument("--interval", type=float, default=60.0)
     ap.add_argument("--deep-every", type=int, default=10,
@@ -17,8 +19,16 @@ def main() -> None:
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
 
-    from .health import NodeHealthAgent
+    from .health import NodeHealthAgent, collect_node_health
 
+    if args.local:
+        import json
+
+        print(json.dumps(collect_node_health(deep=True, sweep=True,
+                                             probe_pairs=args.probe_pairs), indent=2))
+        return
+    if not args.scheduler:
+        ap.error("--scheduler is required (or use --local)")
     agent = NodeHealthAgent(args.scheduler, node_name=args.node_name,
                             interval_s=args.interval, deep_every=args.deep_every,
                             probe_pairs=args.probe_pairs)
