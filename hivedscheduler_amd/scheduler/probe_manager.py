@@ -28,6 +28,10 @@ class ProbeManager:
         # bound: a healthy pair sustains >100 GB/s busbw; use a conservative
         # floor so only truly degraded links trip it)
         self.min_busbw = min_busbw_gbps_per_cell or {2: 50.0, 4: 50.0, 8: 50.0}
+        # daemon hygiene: cap retained groups and expire unpolled tasks so a
+        # long-lived scheduler doesn't grow without bound
+        self.max_groups = 1000
+        self.task_ttl_s = 3600.0
 
     def enqueue_group(self, group_name: str, placements: Dict[str, List[int]]) -> None:
         """placements: node -> leaf cell indices used by the group there."""
@@ -43,8 +47,17 @@ class ProbeManager:
                 })
 
     def poll(self, node: str) -> List[dict]:
-        """Agent polling: returns and drains this node's pending tasks."""
+        """Agent polling: returns and drains this node's pending tasks
+        (expired tasks — e.g. for nodes whose agent never came — are
+        dropped)."""
         with self.lock:
+            now = time.time()
+            for n in list(self.pending):
+                fresh = [t for t in self.pending[n] if now - t["enqueued"] < self.task_ttl_s]
+                if fresh:
+                    self.pending[n] = fresh
+                else:
+                    del self.pending[n]
             return self.pending.pop(node, [])
 
     def report(self, result: dict) -> dict:
@@ -53,6 +66,8 @@ class ProbeManager:
         with self.lock:
             group = result.get("group", "")
             self.results.setdefault(group, []).append(result)
+            while len(self.results) > self.max_groups:
+                self.results.pop(next(iter(self.results)))
             n = len(result.get("leafCellIndices", []))
             floor = self.min_busbw.get(n, 0.0)
             healthy = bool(result.get("ok", False)) and (
