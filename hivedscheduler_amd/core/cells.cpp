@@ -64,12 +64,16 @@ void bindCell(PhysicalCell* pc, VirtualCell* vc) {
 }
 
 // Unbind a physical cell's binding and its ancestors bottom-up while no bound
-// sibling remains; never unbind pinned cells, and never unbind UNHEALTHY
-// ancestors — bad (incl. doomed-bad) cells keep their bindings until healthy
-// (reference hived_algorithm.go:1327-1345). Fuzz-found: unbinding a doomed
-// bad quad here while releaseLeafCell's doomed guard skipped the matching
-// accounting release left the binding and the free list out of sync, and a
-// later re-bind double-allocated the cell.
+// sibling remains; never unbind pinned cells (same guard as the reference,
+// cell_allocation.go:386-420). The complementary unhealthy-cell guard — bad
+// (incl. doomed-bad) cells keep their bindings until healthy, reference
+// hived_algorithm.go:1327-1345 — is applied by the CALLER at leaf level
+// (releaseLeafCell, algorithm.cpp): an unhealthy leaf never reaches this
+// function, so its ancestors' bindings survive too. Fuzz-found rationale for
+// that caller-side guard: unbinding a doomed bad quad here while
+// releaseLeafCell's doomed guard skipped the matching accounting release
+// left the binding and the free list out of sync, and a later re-bind
+// double-allocated the cell.
 void unbindCell(PhysicalCell* c) {
   VirtualCell* boundVirtual = c->virt;
   while (boundVirtual->phys != nullptr && !boundVirtual->phys->pinned) {

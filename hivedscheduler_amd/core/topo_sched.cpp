@@ -211,10 +211,12 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   for (size_t nodeIndex = 0; nodeIndex < cv.size(); nodeIndex++) {
     const NodeView& n = cv[nodeIndex];
     if (n.freeAtPriority >= totalLeafNum) {
-      if (!n.suggested) {
-        *failedReason = "have to use at least one non-suggested node " + n.c->address;
-        return false;
-      }
+      // A non-suggested candidate only disqualifies itself from the
+      // preference, not the request: keep scanning for a suggested
+      // whole-gang host, and if none exists fall through to the greedy
+      // split loop (which alone carries the reference's hard "must use a
+      // non-suggested node" failure, findNodesForPods semantics).
+      if (!n.suggested) continue;
       for (size_t i = 0; i < sortedLeafNums.size(); i++) {
         pickedNodeIndices[i] = static_cast<int>(nodeIndex);
       }

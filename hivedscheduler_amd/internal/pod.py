@@ -181,8 +181,15 @@ def extract_pod_bind_info(pod: Pod) -> PodBindInfo:
     except ValueError:
         data = yaml_load(_convert_old_annotation(raw)) or {}
     else:
-        if any(k in raw for k in ("gpuIsolation", "physicalGpuIndices", "gpuType", "gpuNumber")):
+        if not isinstance(data, dict):
+            # json.loads accepts non-dict documents ("null", a list, a
+            # number); route them through the YAML path's `or {}` guard so
+            # they surface as a 400, not an AttributeError 500
             data = yaml_load(_convert_old_annotation(raw)) or {}
+        elif any(k in raw for k in ("gpuIsolation", "physicalGpuIndices", "gpuType", "gpuNumber")):
+            data = yaml_load(_convert_old_annotation(raw)) or {}
+    if not isinstance(data, dict):
+        data = {}
     return PodBindInfo.from_dict(data)
 
 

@@ -58,6 +58,17 @@ def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False,
     report["mfma_max_err_vs_fp32"] = max_err
     report["mfma_cross_cu_spread"] = tile_spread  # must be exactly 0
     report["mfma_ok"] = bool(tile_spread == 0.0 and max_err < 0.1)
+    # Low-precision pipes (MI355X's headline datapaths): fp8 e4m3 MFMA, the
+    # MX block-scaled fp8 path (K=128), and MX fp4. A GPU whose bf16 pipes
+    # work but whose fp8/fp4 units are broken passes the bf16 check alone.
+    for key, fn in (("mfma_fp8", _mfma_fp8_probe), ("mfma_mx8", _mfma_mx_probe_fp8),
+                    ("mfma_fp4", _mfma_mx_probe_fp4)):
+        err, spread = fn(ops, device)
+        report[f"{key}_max_err_vs_fp32"] = err
+        report[f"{key}_cross_cu_spread"] = spread
+        report[f"{key}_ok"] = bool(spread == 0.0 and err < 0.1)
+    report["mfma_lowprec_ok"] = bool(report["mfma_fp8_ok"] and report["mfma_mx8_ok"]
+                                     and report["mfma_fp4_ok"])
     # CU coverage: a big grid must place waves on every CU of every XCD; a
     # fused-off / hung CU shows up as missing (xcc, se, sh, cu) tuples
     words = ops.cu_coverage(4096).cpu().numpy()
@@ -67,11 +78,63 @@ def gpu_health_report(device: int = 0, quick: bool = True, deep: bool = False,
     expected_cus = report["info"]["multiProcessorCount"]
     report["cu_coverage_ok"] = bool(len(cus) >= expected_cus)
     report["lds_errors"] = int(ops.lds_check(2048, 7))
-    report["healthy"] = bool(report["mfma_ok"] and report["cu_coverage_ok"]
+    report["healthy"] = bool(report["mfma_ok"] and report["mfma_lowprec_ok"]
+                             and report["cu_coverage_ok"]
                              and report["lds_errors"] == 0
                              and report["hbm_gbps"] > 1000.0)
     if deep:
         sweep = ops.hbm_sweep(16 if quick else 64, 4, 1)
         report["hbm_sweep"] = sweep
-        report["healthy"] = bool(report["healthy"] and sweep["errors"] == 0)
+        # A skipped (zero-byte) sweep is not evidence of health, but neither
+        # is it evidence of sickness (a busy tenant GPU can leave too little
+        # free HBM): record it, don't flip `healthy` on it.
+        if not sweep.get("skipped"):
+            report["healthy"] = bool(report["healthy"] and sweep["errors"] == 0)
     return report
+
+
+_FP4_E2M1_VALUES = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+                    -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0]
+
+
+def _mfma_fp8_probe(ops, device):
+    """fp8 e4m3 16x16x32 MFMA vs a torch fp32 reference decoded from the
+    same raw e4m3 bytes. Returns (max_err, cross_cu_spread)."""
+    import torch
+
+    torch.manual_seed(1)
+    A8 = (torch.randn(16, 32) / 8).to(torch.float8_e4m3fn).cuda(device)
+    B8 = (torch.randn(32, 16) / 8).to(torch.float8_e4m3fn).cuda(device)
+    tiles = ops.mfma_check_fp8(A8.view(torch.uint8), B8.view(torch.uint8), 2048)
+    ref = A8.float() @ B8.float()
+    return ((tiles - ref.unsqueeze(0)).abs().max().item(),
+            (tiles - tiles[0].unsqueeze(0)).abs().max().item())
+
+
+def _mfma_mx_probe_fp8(ops, device):
+    """MX block-scaled fp8 (16x16x128 f8f6f4, fmt=0) with unit scales."""
+    import torch
+
+    torch.manual_seed(2)
+    A8 = (torch.randn(16, 128) / 16).to(torch.float8_e4m3fn).cuda(device)
+    B8 = (torch.randn(128, 16) / 16).to(torch.float8_e4m3fn).cuda(device)
+    tiles = ops.mfma_check_mx(A8.view(torch.uint8), B8.view(torch.uint8), 2048, 0)
+    ref = A8.float() @ B8.float()
+    return ((tiles - ref.unsqueeze(0)).abs().max().item(),
+            (tiles - tiles[0].unsqueeze(0)).abs().max().item())
+
+
+def _mfma_mx_probe_fp4(ops, device):
+    """MX block-scaled fp4 (e2m1 codes, fmt=4) with unit scales. Values are
+    exactly representable and the K=128 dot products stay small, so the fp32
+    reference matches bit-for-bit on healthy hardware."""
+    import torch
+
+    g = torch.Generator().manual_seed(3)
+    A4 = torch.randint(0, 16, (16, 128), generator=g, dtype=torch.uint8).cuda(device)
+    B4 = torch.randint(0, 16, (128, 16), generator=g, dtype=torch.uint8).cuda(device)
+    lut = torch.tensor(_FP4_E2M1_VALUES, dtype=torch.float32).cuda(device)
+    tiles = ops.mfma_check_mx(A4, B4, 2048, 4)
+    ref = lut[A4.long()] @ lut[B4.long()]
+    return ((tiles - ref.unsqueeze(0)).abs().max().item(),
+            (tiles - tiles[0].unsqueeze(0)).abs().max().item())

@@ -139,6 +139,137 @@ torch::Tensor mfma_check(torch::Tensor A, torch::Tensor B, long blocks, long rep
 }
 
 // ---------------------------------------------------------------------------
+// FP8 MFMA health check (CDNA4 low-precision pipes). MI355X's headline
+// compute is FP8/FP6/FP4; a GPU whose bf16 matrix pipes work but whose fp8
+// datapath is broken would pass the bf16 check above. Same tile protocol as
+// mfma_check: every wave computes one 16x16 = (16x32 @ 32x16) tile with
+// v_mfma_f32_16x16x32_fp8_fp8 (OCP e4m3), host verifies all tiles bitwise
+// against a torch float8_e4m3fn reference.
+//
+// Fragment layout mirrors the bf16 16x16x32 shape (8 K-elements per lane,
+// here 8 fp8 bytes packed into one i64): lane l holds
+// A[m=l&15][k=(l>>4)*8+j] and B[k=(l>>4)*8+j][n=l&15], j in [0,8);
+// C/D layout is dtype-independent on gfx950.
+// ---------------------------------------------------------------------------
+__global__ void mfma_check_fp8_kernel(const unsigned char* __restrict__ A,
+                                      const unsigned char* __restrict__ B,
+                                      float* __restrict__ C) {
+  int lane = threadIdx.x & 63;
+  int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int m = lane & 15;
+  int kbase = (lane >> 4) * 8;
+  long a = 0, b = 0;
+  for (int j = 0; j < 8; j++) {
+    a |= (long)A[m * 32 + kbase + j] << (8 * j);
+    b |= (long)B[(kbase + j) * 16 + m] << (8 * j);
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc, 0, 0, 0);
+  float* out = C + (size_t)wave * 256;
+  for (int r = 0; r < 4; r++) {
+    int row = (lane >> 4) * 4 + r;
+    out[row * 16 + m] = acc[r];
+  }
+}
+
+// A: [16,32] uint8 (raw OCP e4m3 bytes), B: [32,16] uint8
+// -> [waves, 16, 16] fp32 tiles.
+torch::Tensor mfma_check_fp8(torch::Tensor A, torch::Tensor B, long blocks) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda(), "A/B must be on GPU");
+  TORCH_CHECK(A.dtype() == torch::kUInt8 && B.dtype() == torch::kUInt8,
+              "A/B must be uint8 (raw fp8 e4m3 bytes)");
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 32}) &&
+              B.sizes() == torch::IntArrayRef({32, 16}));
+  A = A.contiguous();
+  B = B.contiguous();
+  int threads = 256;
+  long waves = blocks * (threads / 64);
+  auto C = torch::empty({waves, 16, 16},
+                        torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(mfma_check_fp8_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     A.data_ptr<unsigned char>(), B.data_ptr<unsigned char>(),
+                     C.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return C;
+}
+
+// ---------------------------------------------------------------------------
+// MX block-scaled MFMA health check: v_mfma_f32_16x16x128_f8f6f4, the
+// gfx950-only instruction behind the FP8 (~5 PF) and FP4 (~10 PF) headline
+// rates. fmt selects A/B element format: 0=fp8(e4m3), 4=fp4(e2m1).
+// K=128: lane l holds A[m=l&15][k=(l>>4)*32+j], j in [0,32) — 32 bytes
+// (8 VGPRs) for fp8; for fp4, 32 elements packed 2-per-byte into 16 bytes
+// (low nibble = even k). Scales are e8m0; 0x7F7F7F7F = 1.0 in every byte
+// position so the result is the plain product regardless of scale_sel.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) int i32x8;
+
+template <int fmt>  // immediate operand: FMT must be a compile-time constant
+__global__ void mfma_check_mx_kernel(const unsigned char* __restrict__ A,
+                                     const unsigned char* __restrict__ B,
+                                     float* __restrict__ C) {
+  int lane = threadIdx.x & 63;
+  int wave = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  int m = lane & 15;
+  int kbase = (lane >> 4) * 32;
+  union { i32x8 v; unsigned char b[32]; } a, b;
+  a.v = (i32x8){0, 0, 0, 0, 0, 0, 0, 0};
+  b.v = a.v;
+  if constexpr (fmt == 0) {  // fp8: one byte per element
+    for (int j = 0; j < 32; j++) {
+      a.b[j] = A[m * 128 + kbase + j];
+      b.b[j] = B[(kbase + j) * 16 + m];
+    }
+  } else {  // fp4: two elements per byte, low nibble first
+    for (int j = 0; j < 32; j += 2) {
+      unsigned lo = A[m * 128 + kbase + j] & 0xF;
+      unsigned hi = A[m * 128 + kbase + j + 1] & 0xF;
+      a.b[j / 2] = lo | (hi << 4);
+      lo = B[(kbase + j) * 16 + m] & 0xF;
+      hi = B[(kbase + j + 1) * 16 + m] & 0xF;
+      b.b[j / 2] = lo | (hi << 4);
+    }
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(a.v, b.v, acc, fmt, fmt, 0, 0x7F7F7F7F,
+                                                         0, 0x7F7F7F7F);
+  float* out = C + (size_t)wave * 256;
+  for (int r = 0; r < 4; r++) {
+    int row = (lane >> 4) * 4 + r;
+    out[row * 16 + m] = acc[r];
+  }
+}
+
+// A: [16,128] uint8, B: [128,16] uint8 — raw e4m3 bytes (fmt=0) or fp4 e2m1
+// codes 0..15 one-per-byte (fmt=4). Returns [waves, 16, 16] fp32 tiles.
+torch::Tensor mfma_check_mx(torch::Tensor A, torch::Tensor B, long blocks, long fmt) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda(), "A/B must be on GPU");
+  TORCH_CHECK(A.dtype() == torch::kUInt8 && B.dtype() == torch::kUInt8);
+  TORCH_CHECK(A.sizes() == torch::IntArrayRef({16, 128}) &&
+              B.sizes() == torch::IntArrayRef({128, 16}));
+  TORCH_CHECK(fmt == 0 || fmt == 4, "fmt must be 0 (fp8 e4m3) or 4 (fp4 e2m1)");
+  A = A.contiguous();
+  B = B.contiguous();
+  int threads = 256;
+  long waves = blocks * (threads / 64);
+  auto C = torch::empty({waves, 16, 16},
+                        torch::TensorOptions().dtype(torch::kFloat32).device(A.device()));
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  if (fmt == 0) {
+    hipLaunchKernelGGL((mfma_check_mx_kernel<0>), dim3(blocks), dim3(threads), 0, stream,
+                       A.data_ptr<unsigned char>(), B.data_ptr<unsigned char>(),
+                       C.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL((mfma_check_mx_kernel<4>), dim3(blocks), dim3(threads), 0, stream,
+                       A.data_ptr<unsigned char>(), B.data_ptr<unsigned char>(),
+                       C.data_ptr<float>());
+  }
+  HIP_CHECK(hipGetLastError());
+  return C;
+}
+
+// ---------------------------------------------------------------------------
 // LDS integrity: each workgroup fills a 64 KB LDS slab with address-derived
 // patterns, barriers, and read-verifies from different lanes (catches
 // per-CU LDS faults; HBM and matrix pipes are covered by the other checks).
@@ -172,6 +303,10 @@ long lds_check(long blocks, long seed) {
   hipLaunchKernelGGL(lds_check_kernel, dim3(blocks), dim3(256), 0, stream, errs_d,
                      (unsigned long long)seed);
   HIP_CHECK(hipGetLastError());
+  // The torch stream is non-blocking: a legacy-stream hipMemcpy would not
+  // order against the kernel, racing the error-counter read (a faulty LDS
+  // could read back as 0 errors). Synchronize the launch stream first.
+  HIP_CHECK(hipStreamSynchronize(stream));
   unsigned long long errs = 0;
   HIP_CHECK(hipMemcpy(&errs, errs_d, sizeof(errs), hipMemcpyDeviceToHost));
   HIP_CHECK(hipFree(errs_d));
@@ -343,6 +478,10 @@ py::dict hbm_sweep(long max_gib, long chunk_gib, long seed) {
   d["errors"] = (long long)errs;
   d["write_gbps"] = write_s > 0 ? tested / write_s / 1e9 : 0.0;
   d["verify_gbps"] = verify_s > 0 ? tested / verify_s / 1e9 : 0.0;
+  // A busy tenant GPU can leave < headroom + one chunk free, so the loop
+  // never runs; a zero-byte sweep is NOT evidence of health and callers
+  // (gpu_health_report) must not count it as a pass.
+  d["skipped"] = (bool)(tested == 0);
   return d;
 }
 
@@ -375,6 +514,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Streaming HBM bandwidth in GB/s (triad: 2 reads + 1 write)");
   m.def("mfma_check", &mfma_check, py::arg("A"), py::arg("B"), py::arg("blocks") = 2048,
         py::arg("repeats") = 1, "Per-wave bf16 MFMA tile GEMM across all CUs");
+  m.def("mfma_check_fp8", &mfma_check_fp8, py::arg("A"), py::arg("B"), py::arg("blocks") = 2048,
+        "Per-wave fp8 (OCP e4m3) MFMA tile GEMM across all CUs");
+  m.def("mfma_check_mx", &mfma_check_mx, py::arg("A"), py::arg("B"), py::arg("blocks") = 2048,
+        py::arg("fmt") = 0,
+        "Per-wave MX block-scaled MFMA (16x16x128 f8f6f4) tile; fmt 0=fp8, 4=fp4");
   m.def("p2p_gbps", &p2p_gbps, py::arg("src_dev"), py::arg("dst_dev"), py::arg("size_mb") = 256,
         py::arg("iters") = 10, "xGMI peer-to-peer copy bandwidth in GB/s");
   m.def("hbm_sweep", &hbm_sweep, py::arg("max_gib") = 16, py::arg("chunk_gib") = 4,

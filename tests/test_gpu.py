@@ -51,6 +51,76 @@ def test_mfma_identity():
 
 
 @needs_gpu
+def test_mfma_fp8_numerics_vs_fp32():
+    """fp8 e4m3 MFMA (v_mfma_f32_16x16x32_fp8_fp8) vs a torch fp32 reference
+    decoded from the same raw e4m3 bytes; exercises the CDNA4 low-precision
+    pipes the bf16 check cannot see."""
+    from hivedscheduler_amd.ops import get_ops
+
+    ops = get_ops()
+    torch.manual_seed(11)
+    A8 = (torch.randn(16, 32) / 8).to(torch.float8_e4m3fn).cuda()
+    # asymmetric B catches transposed C-write layouts
+    B8 = ((torch.arange(32 * 16, dtype=torch.float32).reshape(32, 16) / 997 - 0.25)
+          .to(torch.float8_e4m3fn).cuda())
+    tiles = ops.mfma_check_fp8(A8.view(torch.uint8), B8.view(torch.uint8), 64)
+    ref = A8.float() @ B8.float()
+    err = (tiles[0] - ref).abs().max().item()
+    assert err < 0.05, f"fp8 MFMA deviates from fp32 reference: {err}"
+    spread = (tiles - tiles[0].unsqueeze(0)).abs().max().item()
+    assert spread == 0.0, f"cross-CU fp8 MFMA mismatch: {spread}"
+
+
+@needs_gpu
+def test_mfma_mx_fp8_numerics_vs_fp32():
+    """MX block-scaled fp8 (v_mfma_f32_16x16x128_f8f6f4, fmt=0, unit scales)
+    vs torch fp32 — the instruction behind the ~5 PF fp8 headline rate."""
+    from hivedscheduler_amd.ops import get_ops
+
+    ops = get_ops()
+    torch.manual_seed(12)
+    A8 = (torch.randn(16, 128) / 16).to(torch.float8_e4m3fn).cuda()
+    B8 = ((torch.arange(128 * 16, dtype=torch.float32).reshape(128, 16) / 4093 - 0.25)
+          .to(torch.float8_e4m3fn).cuda())
+    tiles = ops.mfma_check_mx(A8.view(torch.uint8), B8.view(torch.uint8), 64, 0)
+    ref = A8.float() @ B8.float()
+    err = (tiles[0] - ref).abs().max().item()
+    assert err < 0.05, f"MX fp8 MFMA deviates from fp32 reference: {err}"
+    spread = (tiles - tiles[0].unsqueeze(0)).abs().max().item()
+    assert spread == 0.0, f"cross-CU MX fp8 MFMA mismatch: {spread}"
+
+
+@needs_gpu
+def test_mfma_mx_fp4_numerics_exact():
+    """MX fp4 (e2m1 codes, fmt=4, unit scales): all products are multiples of
+    0.25 with sums << 2^24, so the fp32 reference must match bit-for-bit."""
+    from hivedscheduler_amd.ops import get_ops, _FP4_E2M1_VALUES
+
+    ops = get_ops()
+    g = torch.Generator().manual_seed(13)
+    A4 = torch.randint(0, 16, (16, 128), generator=g, dtype=torch.uint8).cuda()
+    B4 = torch.randint(0, 16, (128, 16), generator=g, dtype=torch.uint8).cuda()
+    lut = torch.tensor(_FP4_E2M1_VALUES, dtype=torch.float32).cuda()
+    tiles = ops.mfma_check_mx(A4, B4, 64, 4)
+    ref = lut[A4.long()] @ lut[B4.long()]
+    assert torch.equal(tiles[0], ref), (
+        f"MX fp4 MFMA mismatch: max err {(tiles[0] - ref).abs().max().item()}")
+    spread = (tiles - tiles[0].unsqueeze(0)).abs().max().item()
+    assert spread == 0.0, f"cross-CU MX fp4 MFMA mismatch: {spread}"
+
+
+@needs_gpu
+def test_health_report_includes_lowprec():
+    from hivedscheduler_amd.ops import gpu_health_report
+
+    rep = gpu_health_report(0, quick=True)
+    assert rep["mfma_fp8_ok"] and rep["mfma_mx8_ok"] and rep["mfma_fp4_ok"], rep
+    assert rep["mfma_fp8_cross_cu_spread"] == 0.0
+    assert rep["mfma_fp4_cross_cu_spread"] == 0.0
+    assert rep["mfma_lowprec_ok"] and rep["healthy"]
+
+
+@needs_gpu
 def test_hbm_bandwidth_sane():
     from hivedscheduler_amd.ops import get_ops
 
