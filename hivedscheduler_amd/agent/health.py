@@ -97,19 +97,61 @@ def collect_node_health(
         runner = CellProbeRunner(min_busbw_gbps=min_pair_busbw_gbps)
         if runner.available():
             report["pairs"] = {}
+            links = []
             for p in range(n // 2):
                 pair = [2 * p, 2 * p + 1]
                 try:
                     res = runner.probe_cell(pair, size_mb=64, iters=10)
                     report["pairs"][f"{pair[0]}-{pair[1]}"] = res
-                    if res.get("ok") and not res.get("healthy", True):
-                        # degraded xGMI link: mark both endpoints' leaves bad
-                        for g in pair:
-                            report["gpus"][str(g)]["healthy"] = False
-                            report["gpus"][str(g)]["xgmi_degraded"] = True
+                    if res.get("ok"):
+                        # first-class LINK report: a degraded xGMI link marks
+                        # the link (scheduler avoids co-placing the endpoints)
+                        # while both GPUs stay schedulable for 1-GPU work
+                        links.append({"a": pair[0], "b": pair[1],
+                                      "healthy": bool(res.get("healthy", True)),
+                                      "gbps": float(res.get("busbw_gbps") or 0.0)})
                 except Exception as e:
                     report["pairs"][f"{pair[0]}-{pair[1]}"] = {"ok": False, "error": str(e)[:200]}
+            if links:
+                report["links"] = links
     return report
+
+
+def p2p_link_matrix(n: Optional[int] = None, size_mb: int = 64, iters: int = 5,
+                    min_link_gbps: float = 40.0) -> dict:
+    """Measure the full xGMI p2p bandwidth matrix with the HIP copy kernel
+    (ops.p2p_gbps) and derive per-link health. This is the cheapest way to
+    localize a degraded link to ONE pair when a >=4-GPU collective probe
+    reads low: every GPU pair on an 8x MI355X node is directly connected
+    (7 links x ~153 GB/s per direction), so matrix[i][j] well below the
+    floor indicts exactly link i<->j.
+
+    Returns {"matrix": {"i-j": gbps}, "links": [{a, b, gbps, healthy}]}
+    shaped for the scheduler's /v1/health/nodes intake.
+    """
+    import torch
+
+    from ..ops import get_ops
+
+    ops = get_ops()
+    count = n if n is not None else (torch.cuda.device_count() if torch.cuda.is_available() else 0)
+    out: dict = {"matrix": {}, "links": []}
+    for i in range(count):
+        for j in range(i + 1, count):
+            try:
+                # per-direction copies; a link is as sick as its worse direction
+                fwd = ops.p2p_gbps(i, j, size_mb, iters)
+                rev = ops.p2p_gbps(j, i, size_mb, iters)
+                gbps = min(fwd, rev)
+            except Exception as e:
+                log.warning("p2p probe %d<->%d failed: %s", i, j, e)
+                out["links"].append({"a": i, "b": j, "healthy": False, "gbps": 0.0,
+                                     "error": str(e)[:200]})
+                continue
+            out["matrix"][f"{i}-{j}"] = round(gbps, 1)
+            out["links"].append({"a": i, "b": j, "healthy": bool(gbps >= min_link_gbps),
+                                 "gbps": round(gbps, 1)})
+    return out
 
 
 class NodeHealthAgent:
@@ -117,14 +159,19 @@ class NodeHealthAgent:
 
     def __init__(self, scheduler_url: str, node_name: Optional[str] = None,
                  interval_s: float = 60.0, deep_every: int = 10,
-                 sweep_every: int = 60, probe_pairs: bool = False):
+                 sweep_every: int = 60, probe_pairs: bool = True,
+                 p2p_matrix_every: int = 30):
         self.scheduler_url = scheduler_url.rstrip("/")
         self.node_name = node_name or socket.gethostname()
         self.interval_s = interval_s
         self.deep_every = deep_every
         # stuck-bit HBM sweep cadence (GPU must be idle-ish; 16 GiB ~ 5 s/GPU)
         self.sweep_every = sweep_every
+        # pair probes default ON: the xGMI checks are the point of the agent
         self.probe_pairs = probe_pairs
+        # full p2p matrix cadence (28 pairs x 2 directions, ~1 min on 8 GPUs):
+        # localizes a degraded link to one pair when collective probes read low
+        self.p2p_matrix_every = p2p_matrix_every
         self._rounds = 0
 
     def post_report(self, report: dict) -> bool:
@@ -171,10 +218,20 @@ class NodeHealthAgent:
     def run_once(self) -> dict:
         deep = (self._rounds % self.deep_every) == 0
         sweep = (self._rounds % self.sweep_every) == 0
+        matrix = (self._rounds % self.p2p_matrix_every) == 0
         self._rounds += 1
         report = collect_node_health(deep=deep, probe_pairs=self.probe_pairs and deep,
                                      sweep=sweep and deep)
         report["node"] = self.node_name
+        if matrix and deep:
+            try:
+                m = p2p_link_matrix()
+                if m["links"]:
+                    report["p2p_matrix"] = m["matrix"]
+                    # matrix verdicts override pair-probe verdicts (finer)
+                    report["links"] = m["links"]
+            except Exception as e:
+                log.warning("p2p matrix sweep failed: %s", e)
         self.post_report(report)
         report["placement_probes"] = self.run_placement_probes()
         return report

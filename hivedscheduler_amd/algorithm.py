@@ -117,6 +117,18 @@ class HivedAlgorithm:
         pair/quad/node cells. Independent of node-level health."""
         self._call(self._core.set_leaf_cell_healthy, node, leaf_index, healthy)
 
+    def set_xgmi_link_healthy(self, node: str, a: int, b: int, healthy: bool,
+                              gbps: float = 0.0) -> None:
+        """First-class xGMI link health: a degraded link between GPUs a and b
+        of one node makes multi-GPU placements avoid co-placing the two
+        endpoints, while both GPUs stay schedulable for 1-GPU work (unlike
+        set_leaf_cell_healthy, which removes a GPU entirely)."""
+        self._call(self._core.set_xgmi_link_healthy, node, a, b, healthy, gbps)
+
+    def get_xgmi_links(self, node: str) -> List[dict]:
+        """Per-node link table: [{a, b, gbps, healthy}, ...]."""
+        return self._call(self._core.xgmi_links, node)
+
     def all_nodes(self) -> List[str]:
         return self._call(self._core.all_nodes)
 

@@ -64,6 +64,11 @@ class PhysicalCellSpec:
     cellType: CellType = ""
     cellAddress: str = ""
     pinnedCellId: str = ""
+    # MI355X extensions (rocm-topo-discover output): measured per-GPU HBM
+    # bytes (leaf entries) and the measured xGMI link table
+    # [{a, b, gbps, healthy}] (node-level entries)
+    hbmBytes: int = 0
+    xgmiLinks: List[Dict[str, Any]] = field(default_factory=list)
     cellChildren: List["PhysicalCellSpec"] = field(default_factory=list)
 
     @staticmethod
@@ -72,6 +77,8 @@ class PhysicalCellSpec:
             cellType=str(d.get("cellType", "") or ""),
             cellAddress=str(d.get("cellAddress", "") if d.get("cellAddress") is not None else ""),
             pinnedCellId=str(d.get("pinnedCellId", "") or ""),
+            hbmBytes=int(d.get("hbmBytes") or 0),
+            xgmiLinks=list(d.get("xgmiLinks") or []),
             cellChildren=[PhysicalCellSpec.from_dict(c) for c in (d.get("cellChildren") or [])],
         )
 
@@ -79,6 +86,10 @@ class PhysicalCellSpec:
         out: Dict[str, Any] = {"cellType": self.cellType, "cellAddress": self.cellAddress}
         if self.pinnedCellId:
             out["pinnedCellId"] = self.pinnedCellId
+        if self.hbmBytes:
+            out["hbmBytes"] = self.hbmBytes
+        if self.xgmiLinks:
+            out["xgmiLinks"] = [dict(l) for l in self.xgmiLinks]
         if self.cellChildren:
             out["cellChildren"] = [c.to_dict() for c in self.cellChildren]
         return out
@@ -177,6 +188,10 @@ class PodSchedulingSpec:
     gangReleaseEnable: bool = False
     lazyPreemptionEnable: bool = False
     ignoreK8sSuggestedNodes: bool = True
+    # MI355X extension: minimum measured HBM per leaf cell in bytes (0 = any).
+    # Leaves whose discovered/agent-measured capacity falls short (a GPU
+    # reporting < 288 GB is sick) are avoided for this request.
+    hbmBytesPerCell: int = 0
     affinityGroup: Optional[AffinityGroupSpec] = None
 
     @staticmethod
@@ -194,6 +209,7 @@ class PodSchedulingSpec:
             gangReleaseEnable=bool(d.get("gangReleaseEnable", False)),
             lazyPreemptionEnable=bool(d.get("lazyPreemptionEnable", False)),
             ignoreK8sSuggestedNodes=bool(d.get("ignoreK8sSuggestedNodes", True)),
+            hbmBytesPerCell=int(d.get("hbmBytesPerCell") or 0),
             affinityGroup=AffinityGroupSpec.from_dict(ag) if ag else None,
         )
 
@@ -206,6 +222,8 @@ class PodSchedulingSpec:
             "lazyPreemptionEnable": self.lazyPreemptionEnable,
             "ignoreK8sSuggestedNodes": self.ignoreK8sSuggestedNodes,
         }
+        if self.hbmBytesPerCell:
+            out["hbmBytesPerCell"] = self.hbmBytesPerCell
         if self.pinnedCellId:
             out["pinnedCellId"] = self.pinnedCellId
         if self.leafCellType:

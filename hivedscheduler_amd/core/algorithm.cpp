@@ -336,6 +336,7 @@ bool HivedCore::scheduleNewAffinityGroup(const PodSpec& s,
   sr.podLeafCellNums = s.groupPodNums;
   sr.suggestedNodes = &suggestedNodes;
   sr.ignoreSuggestedNodes = s.ignoreK8sSuggestedNodes;
+  sr.hbmBytes = s.hbmBytesPerCell;
   validateSchedulingRequest(sr, podKey);
   if (!sr.pinnedCellId.empty()) {
     sr.chain = pinnedPhysical_[s.vc][s.pinnedCellId]->chain;
@@ -466,10 +467,32 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& sr, Placement<P
     }
   }
 
-  std::map<int, int> freeCellNumCopy = allVCFreeCellNum_[sr.chain];
-  if (mapVirtualPlacementToPhysical(preassigned, nonPreassigned,
-                                    freeCellList_[sr.chain].shallowCopy(), freeCellNumCopy,
-                                    *sr.suggestedNodes, sr.ignoreSuggestedNodes, bindings)) {
+  // Two-tier mapping when the chain carries degraded xGMI links: first
+  // demand a link-clean mapping (no two gang leaves across a degraded link;
+  // backtracking explores alternatives), then accept a dirty one — capacity
+  // guarantees outrank link quality. Each attempt gets fresh free-list /
+  // accounting copies (the mapper mutates them).
+  bool anyBadLinks = false;
+  {
+    auto& ccl = fullCellList_[sr.chain];
+    for (Cell* c : ccl.at(ccl.top())) {
+      if (static_cast<PhysicalCell*>(c)->badLinksUnder > 0) {
+        anyBadLinks = true;
+        break;
+      }
+    }
+  }
+  const std::unordered_map<VirtualCell*, PhysicalCell*> seedBindings = bindings;
+  auto tryMap = [&](bool honorLinks) {
+    bindings = seedBindings;  // drop partial picks from a failed attempt
+    std::map<int, int> freeCellNumCopy = allVCFreeCellNum_[sr.chain];
+    return mapVirtualPlacementToPhysical(preassigned, nonPreassigned,
+                                         freeCellList_[sr.chain].shallowCopy(), freeCellNumCopy,
+                                         *sr.suggestedNodes, sr.ignoreSuggestedNodes, bindings,
+                                         sr.hbmBytes, honorLinks);
+  };
+  bool mapped = anyBadLinks ? (tryMap(true) || tryMap(false)) : tryMap(false);
+  if (mapped) {
     phys->clear();
     for (auto& [leafNum, pods] : *virt) {
       for (auto& pod : pods) {
@@ -514,7 +537,7 @@ bool HivedCore::scheduleOpportunisticGroup(const SchedulingRequest& sr,
   Placement<Cell> generic;
   if (!opportunisticSchedulers_.at(sr.chain).Schedule(sr.podLeafCellNums, kOpportunisticPriority,
                                                       *sr.suggestedNodes, sr.ignoreSuggestedNodes,
-                                                      &generic, failedReason)) {
+                                                      &generic, failedReason, sr.hbmBytes)) {
     *failedReason += " when scheduling in physical cluster";
     return false;
   }

@@ -27,7 +27,23 @@ struct AllocCtx {
   // free cells — other VCs' guarantees — survive. Backtracking may leave
   // stale entries; that only biases packing, never correctness.
   std::unordered_set<PhysicalCell*>& claimed;
+  // minimum measured HBM per leaf (0 = any): leaves below are unusable
+  long long minHbm = 0;
+  // when true, a leaf candidate whose degraded-link peer is already claimed
+  // by this mapping round is unusable (the gang would all-reduce over the
+  // sick link); the backtracking search then finds a clean mapping or fails,
+  // and the caller retries with honorLinks=false (capacity over quality)
+  bool honorLinks = false;
 };
+
+// Number of gang leaves that will land under this vertex's physical binding
+// (vertex-tree leaves are the placed level-1 cells).
+int vertexLeafDemand(const BindingVertex* v) {
+  if (v->children.empty()) return v->cell->totalLeaf;
+  int n = 0;
+  for (auto& ch : v->children) n += vertexLeafDemand(ch.get());
+  return n;
+}
 
 void markClaimed(const AllocCtx& ctx, PhysicalCell* c) {
   for (PhysicalCell* a = c; a != nullptr; a = static_cast<PhysicalCell*>(a->parent)) {
@@ -35,16 +51,32 @@ void markClaimed(const AllocCtx& ctx, PhysicalCell* c) {
   }
 }
 
+// honorLinks hard constraint, checked at PICK time (claims accumulate while
+// the backtracking search runs): a leaf whose degraded-link peer is already
+// claimed by this round would put the gang's collective on the sick link.
+bool leafLinkConflict(const AllocCtx& ctx, PhysicalCell* leaf) {
+  if (!ctx.honorLinks || leaf->badLinkPeers.empty()) return false;
+  for (PhysicalCell* peer : leaf->badLinkPeers) {
+    if (ctx.claimed.count(peer)) return true;
+  }
+  return false;
+}
+
 // Usable = unbound, not a known-bad single-node cell, and (unless ignoring
 // suggestions) at least one node within the suggested set. Sorted by
 // opportunistic usage ascending to minimize preemption of opportunistic pods.
 std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& candidates,
-                                                  int numNeeded, const AllocCtx& ctx) {
+                                                  int numNeeded, const AllocCtx& ctx,
+                                                  int demandLeaves = 0) {
   std::vector<PhysicalCell*> usable;
   for (Cell* cc : candidates) {
     auto* c = static_cast<PhysicalCell*>(cc);
     if (c->virt != nullptr) continue;
     if (c->nodes.size() == 1 && !c->healthy) continue;
+    if (ctx.minHbm > 0 && c->level == kLowestLevel && c->hbmBytes > 0 &&
+        c->hbmBytes < ctx.minHbm) {
+      continue;  // leaf's measured HBM falls short of the request's demand
+    }
     if (!ctx.ignoreSuggested) {
       bool anySuggested = false;
       for (auto& n : c->nodes) {
@@ -86,12 +118,22 @@ std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& cand
   auto parentClaimed = [&ctx](PhysicalCell* c) {
     return c->parent != nullptr && ctx.claimed.count(static_cast<PhysicalCell*>(c->parent)) > 0;
   };
+  // xGMI link preference (weakest key, after the safety-driven packing keys
+  // and the opportunistic-disruption key): a multi-leaf vertex avoids
+  // candidates containing degraded links (its leaves would all-reduce over
+  // the sick link); a single-leaf vertex PREFERS them (parking 1-GPU work on
+  // degraded pairs keeps clean pairs free for gangs).
+  auto linkKey = [demandLeaves](PhysicalCell* c) {
+    return demandLeaves >= 2 ? c->badLinksUnder : -c->badLinksUnder;
+  };
   std::stable_sort(usable.begin(), usable.end(), [&](PhysicalCell* a, PhysicalCell* b) {
     bool pa = parentClaimed(a), pb = parentClaimed(b);
     if (pa != pb) return pa;
     int fa = freeBuddies(a), fb = freeBuddies(b);
     if (fa != fb) return fa < fb;
-    return a->usedAt(kOpportunisticPriority) < b->usedAt(kOpportunisticPriority);
+    int oa = a->usedAt(kOpportunisticPriority), ob = b->usedAt(kOpportunisticPriority);
+    if (oa != ob) return oa < ob;
+    return demandLeaves != 0 && linkKey(a) < linkKey(b);
   });
   if (mapDebug()) {
     fprintf(stderr, "[sort]");
@@ -105,11 +147,26 @@ std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& cand
 
 // Backtracking bipartite match of virtual cells onto candidate physical cells,
 // recursing into children to preserve the intra-cell topology.
-bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cells,
+bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cellsIn,
                                const std::vector<Cell*>& candidatesIn, const AllocCtx& ctx,
                                bool returnPicked, std::vector<PhysicalCell*>* picked) {
+  // Multi-leaf vertices map first (demand descending) so they get the
+  // link-clean candidates and single-leaf vertices take the leftovers
+  // (incl. degraded pairs, which suit them fine).
+  std::vector<BindingVertex*> cells = cellsIn;
+  int maxDemand = 0;
+  {
+    std::vector<int> demand(cells.size());
+    for (size_t i = 0; i < cells.size(); i++) {
+      demand[i] = vertexLeafDemand(cells[i]);
+      maxDemand = std::max(maxDemand, demand[i]);
+    }
+    std::stable_sort(cells.begin(), cells.end(), [&](BindingVertex* a, BindingVertex* b) {
+      return vertexLeafDemand(a) > vertexLeafDemand(b);
+    });
+  }
   std::vector<PhysicalCell*> candidates =
-      getUsablePhysicalCells(candidatesIn, static_cast<int>(cells.size()), ctx);
+      getUsablePhysicalCells(candidatesIn, static_cast<int>(cells.size()), ctx, maxDemand);
   if (candidates.empty() && !cells.empty()) return false;
   int n = static_cast<int>(cells.size());
   int m = static_cast<int>(candidates.size());
@@ -123,6 +180,7 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cells,
       PhysicalCell* candidate = candidates[candidateIndex];
       bool ok;
       if (candidate->level == kLowestLevel) {
+        if (leafLinkConflict(ctx, candidate)) continue;
         ok = true;
         ctx.bindings[cells[cellIndex]->cell] = candidate;
         markClaimed(ctx, candidate);
@@ -181,7 +239,8 @@ bool buddyAlloc(BindingVertex* cell, ChainCellList& freeList, int currentLevel,
     if (mapDebug()) fprintf(stderr, "[map]  -> no usable candidate at own level\n");
     return false;
   }
-  std::vector<PhysicalCell*> freeCells = getUsablePhysicalCells(freeList.at(currentLevel), 1, ctx);
+  std::vector<PhysicalCell*> freeCells =
+      getUsablePhysicalCells(freeList.at(currentLevel), 1, ctx, vertexLeafDemand(cell));
   if (freeCells.empty()) return false;
   for (PhysicalCell* c : freeCells) {
     auto saved = freeList.at(currentLevel - 1);
@@ -267,9 +326,19 @@ bool HivedCore::mapVirtualPlacementToPhysical(
     std::vector<BindingVertex*>& preassigned, std::vector<std::vector<BindingVertex*>>& nonPreassigned,
     ChainCellList freeList, std::map<int, int> freeCellNum,
     const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
-    std::unordered_map<VirtualCell*, PhysicalCell*>& bindings) {
+    std::unordered_map<VirtualCell*, PhysicalCell*>& bindings, long long minHbmBytes,
+    bool honorLinks) {
   std::unordered_set<PhysicalCell*> claimed;
-  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings, claimed};
+  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings, claimed, minHbmBytes, honorLinks};
+  // pre-claim physical leaves ALREADY bound for this placement (the caller
+  // seeds `bindings` with bound virtual leaves, which skip the vertex trees)
+  // so the link constraint sees the whole gang, not just the new picks
+  if (honorLinks) {
+    for (auto& [v, p] : bindings) {
+      (void)v;
+      if (p != nullptr) markClaimed(ctx, p);
+    }
+  }
   for (BindingVertex* c : preassigned) {
     if (!buddyAlloc(c, freeList, getLowestFreeCellLevel(freeList, c->cell->level), ctx)) {
       if (!safeRelaxedBuddyAlloc(c, freeList, freeCellNum, c->cell->level, ctx)) {
@@ -534,6 +603,67 @@ void HivedCore::setLeafCellHealthy(const std::string& node, int leafIndex, bool 
       return;
     }
   }
+}
+
+// First-class xGMI link health (BASELINE north star). A degraded link is
+// recorded on the LINK, not the endpoint leaves: badLinksUnder is rolled up
+// from the endpoints' LCA so pair/quad/node cells know a degraded link lies
+// under them, while both endpoint GPUs stay schedulable for 1-GPU work and
+// for placements that do not co-place the two endpoints. This extends the
+// reference's leaf-only healthiness seam (cell.go:302-312) with a per-link
+// dimension the reference cannot express.
+void HivedCore::setXgmiLinkHealthy(const std::string& node, int a, int b, bool healthy,
+                                   double gbps) {
+  if (a == b) throw HivedError::BadRequest("xGMI link endpoints must differ: " + node);
+  if (a > b) std::swap(a, b);
+  auto it = nodeLeafCellsStorage_.find(node);
+  if (it == nodeLeafCellsStorage_.end()) return;
+  PhysicalCell* la = nullptr;
+  PhysicalCell* lb = nullptr;
+  for (PhysicalCell* leaf : it->second) {
+    if (leaf->leafIndices.empty()) continue;
+    if (leaf->leafIndices[0] == a) la = leaf;
+    if (leaf->leafIndices[0] == b) lb = leaf;
+  }
+  if (la == nullptr || lb == nullptr || la->chain != lb->chain) return;
+  XgmiLink& rec = xgmiLinks_[node][{a, b}];
+  bool wasHealthy = rec.a == nullptr ? true : rec.healthy;
+  rec.a = la;
+  rec.b = lb;
+  if (gbps > 0) rec.gbps = gbps;
+  rec.healthy = healthy;
+  if (healthy == wasHealthy) return;
+  if (healthy) {
+    auto erasePeer = [](PhysicalCell* from, PhysicalCell* peer) {
+      auto& v = from->badLinkPeers;
+      v.erase(std::remove(v.begin(), v.end(), peer), v.end());
+    };
+    erasePeer(la, lb);
+    erasePeer(lb, la);
+  } else {
+    la->badLinkPeers.push_back(lb);
+    lb->badLinkPeers.push_back(la);
+  }
+  // roll the count up from the endpoints' LCA (both are leaves: walk in step)
+  Cell* x = la;
+  Cell* y = lb;
+  while (x != y && x != nullptr && y != nullptr) {
+    x = x->parent;
+    y = y->parent;
+  }
+  int delta = healthy ? -1 : 1;
+  for (Cell* c = x; c != nullptr; c = c->parent) {
+    static_cast<PhysicalCell*>(c)->badLinksUnder += delta;
+  }
+}
+
+std::vector<std::tuple<int, int, double, bool>> HivedCore::xgmiLinks(
+    const std::string& node) const {
+  std::vector<std::tuple<int, int, double, bool>> out;
+  auto it = xgmiLinks_.find(node);
+  if (it == xgmiLinks_.end()) return out;
+  for (auto& [key, l] : it->second) out.emplace_back(key.first, key.second, l.gbps, l.healthy);
+  return out;
 }
 
 // A cell is bad if ANY child is bad; propagate from leaf up.

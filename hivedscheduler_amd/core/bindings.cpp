@@ -49,6 +49,19 @@ ClusterSpec parseClusterSpec(const py::dict& d) {
     c.type = dstr(cd, "cellType");
     c.address = dstr(cd, "cellAddress");
     c.pinnedId = dstr(cd, "pinnedCellId");
+    if (cd.contains("hbmBytes") && !cd["hbmBytes"].is_none()) {
+      c.hbmBytes = py::cast<long long>(cd["hbmBytes"]);
+    }
+    // discovery-measured xGMI link table (node-level entries):
+    // [{a, b, gbps, healthy}, ...]
+    if (cd.contains("xgmiLinks") && !cd["xgmiLinks"].is_none()) {
+      for (auto lo : py::cast<py::list>(cd["xgmiLinks"])) {
+        py::dict ld = py::cast<py::dict>(lo);
+        double gbps = ld.contains("gbps") && !ld["gbps"].is_none() ? py::cast<double>(ld["gbps"]) : 0.0;
+        c.xgmiLinks.emplace_back(static_cast<int>(dint(ld, "a")), static_cast<int>(dint(ld, "b")),
+                                 gbps, dbool(ld, "healthy", true));
+      }
+    }
     if (cd.contains("cellChildren") && !cd["cellChildren"].is_none()) {
       for (auto child : py::cast<py::list>(cd["cellChildren"])) {
         c.children.push_back(parseCell(py::cast<py::dict>(child)));
@@ -96,6 +109,12 @@ PodSpec parsePodSpec(const py::dict& d) {
   s.gangReleaseEnable = dbool(d, "gangReleaseEnable");
   s.lazyPreemptionEnable = dbool(d, "lazyPreemptionEnable");
   s.ignoreK8sSuggestedNodes = dbool(d, "ignoreK8sSuggestedNodes", true);
+  if (d.contains("hbmBytesPerCell") && !d["hbmBytesPerCell"].is_none()) {
+    s.hbmBytesPerCell = py::cast<long long>(d["hbmBytesPerCell"]);
+    if (s.hbmBytesPerCell < 0) {
+      throw HivedError::BadRequest("hbmBytesPerCell must be non-negative");
+    }
+  }
   if (d.contains("affinityGroup") && !d["affinityGroup"].is_none()) {
     py::dict ag = py::cast<py::dict>(d["affinityGroup"]);
     s.groupName = dstr(ag, "name");
@@ -190,6 +209,11 @@ py::dict physicalCellStatus(PhysicalCell* c, bool withChildren = true) {
     d["physicalNode"] = c->nodes.empty() ? "" : c->nodes[0];
     d["leafCellIndex"] = c->leafIndices.empty() ? -1 : c->leafIndices[0];
     d["hbmBytes"] = c->hbmBytes;
+  } else if (c->badLinksUnder > 0) {
+    // first-class xGMI link state: degraded links under this pair/quad/node
+    // (the cell stays Healthy — its GPUs work — but multi-GPU placements
+    // avoid co-placing the degraded link's endpoints)
+    d["badXgmiLinksUnder"] = c->badLinksUnder;
   }
   if (withChildren && !c->children.empty()) {
     py::list children;
@@ -286,6 +310,21 @@ class PyHivedCore {
   void setNodeHealthy(const std::string& node, bool healthy) { core_.setNodeHealthy(node, healthy); }
   void setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy) {
     core_.setLeafCellHealthy(node, leafIndex, healthy);
+  }
+  void setXgmiLinkHealthy(const std::string& node, int a, int b, bool healthy, double gbps) {
+    core_.setXgmiLinkHealthy(node, a, b, healthy, gbps);
+  }
+  py::list xgmiLinks(const std::string& node) const {
+    py::list out;
+    for (auto& [a, b, gbps, healthy] : core_.xgmiLinks(node)) {
+      py::dict d;
+      d["a"] = a;
+      d["b"] = b;
+      d["gbps"] = gbps;
+      d["healthy"] = healthy;
+      out.append(d);
+    }
+    return out;
   }
   std::vector<std::string> allNodes() const { return core_.allNodes(); }
   std::vector<std::string> badNodes() const {
@@ -488,6 +527,9 @@ PYBIND11_MODULE(hivedcore, m) {
   py::class_<PyHivedCore>(m, "HivedCore")
       .def(py::init<const py::dict&>(), py::arg("spec"))
       .def("set_node_healthy", &PyHivedCore::setNodeHealthy, py::arg("node"), py::arg("healthy"))
+      .def("set_xgmi_link_healthy", &PyHivedCore::setXgmiLinkHealthy, py::arg("node"),
+           py::arg("a"), py::arg("b"), py::arg("healthy"), py::arg("gbps") = 0.0)
+      .def("xgmi_links", &PyHivedCore::xgmiLinks, py::arg("node"))
       .def("set_leaf_cell_healthy", &PyHivedCore::setLeafCellHealthy, py::arg("node"),
            py::arg("leaf_index"), py::arg("healthy"))
       .def("all_nodes", &PyHivedCore::allNodes)

@@ -55,6 +55,8 @@ struct BuildContext {
   std::map<std::string, ChainMeta> chainMeta;
   std::map<std::string, std::vector<PhysicalCell*>> nodeLeafCells;  // node -> leaf cells
   std::map<std::string, PhysicalCell*> pinnedCellsById;
+  // (node, (a, b, gbps, healthy)) link specs deferred until leaf maps exist
+  std::vector<std::pair<std::string, std::tuple<int, int, double, bool>>> pendingLinks;
 
   BuildContext(HivedCore* h_, const ClusterSpec& s) : h(h_), spec(s) {}
 
@@ -79,6 +81,11 @@ struct BuildContext {
         throw HivedError::BadRequest("pinnedCellId " + cs.pinnedId + " used by multiple physical cells");
       }
     }
+    // discovery-measured xGMI link table (node-level entries): applied after
+    // construction, once nodeLeafCellsStorage_ is populated
+    if (!cs.xgmiLinks.empty()) {
+      for (auto& l : cs.xgmiLinks) pendingLinks.emplace_back(ownNodeName, l);
+    }
     if (level == kLowestLevel) {
       if (ownNodeName.empty()) {
         // chain without a node level: the leaf's own address is the node name
@@ -94,7 +101,10 @@ struct BuildContext {
       c->totalLeaf = 1;
       c->nodes = {ownNodeName};
       c->leafIndices = {leafIndex};
-      c->hbmBytes = 288LL * 1024 * 1024 * 1024;
+      // measured per-GPU capacity from discovery when provided (a GPU
+      // reporting less than the nominal 288 GB is a health signal and is
+      // avoided by requests carrying hbmBytesPerCell), nominal otherwise
+      c->hbmBytes = cs.hbmBytes > 0 ? cs.hbmBytes : 288LL * 1024 * 1024 * 1024;
       nodeLeafCells[ownNodeName].push_back(c);
     } else {
       int expected = childCountAt(spec.cellTypes, c->typeName);
@@ -265,6 +275,11 @@ void HivedCore::buildFromSpec(const ClusterSpec& spec) {
   ctx.build();
   nodeLeafCellsStorage_.clear();
   for (auto& [node, cells] : ctx.nodeLeafCells) nodeLeafCellsStorage_[node] = cells;
+  // apply the discovery-measured xGMI link table (records gbps; a link
+  // reported unhealthy at config time degrades its pair/quad from the start)
+  for (auto& [node, l] : ctx.pendingLinks) {
+    setXgmiLinkHealthy(node, std::get<0>(l), std::get<1>(l), std::get<3>(l), std::get<2>(l));
+  }
 }
 
 // Validates VC quota against the physical cluster and initializes the

@@ -97,9 +97,28 @@ struct PhysicalCell : Cell {
   Group* usingGroup = nullptr;       // Allocated / BeingPreempted group (leaf only)
   Group* reservingGroup = nullptr;   // Preempting group (leaf only)
   std::string otVC;                  // VC using this cell opportunistically
-  // MI355X hardware attributes
+  // MI355X hardware attributes (BASELINE north star: 288 GB HBM and
+  // xGMI-link health as first-class cell attributes)
   long long hbmBytes = 0;
+  // count of degraded xGMI links whose BOTH endpoints are leaves under this
+  // cell (leaf cells: 0). Rolled up from setXgmiLinkHealthy: the LCA of the
+  // link's endpoints and all its ancestors carry the count. A link-degraded
+  // pair/quad stays usable for placements that avoid co-placing the two
+  // endpoints (and for 1-GPU work) — unlike leaf badness, which removes the
+  // GPU entirely. Extends the reference's healthiness seam (cell.go:302-312)
+  // with a per-link dimension.
+  int badLinksUnder = 0;
+  // leaf cells only: peer leaves connected by a currently-degraded link
+  std::vector<PhysicalCell*> badLinkPeers;
   bool isPhysical() const override { return true; }
+};
+
+// One xGMI link's measured state (node-local, endpoints are GPU indices).
+struct XgmiLink {
+  PhysicalCell* a = nullptr;  // leaf cells, a->leafIndices[0] < b->leafIndices[0]
+  PhysicalCell* b = nullptr;
+  double gbps = 0.0;          // measured bandwidth (0 = not measured)
+  bool healthy = true;
 };
 
 struct VirtualCell : Cell {
@@ -192,6 +211,11 @@ struct PhysCellSpec {
   std::string address;
   std::string pinnedId;
   std::vector<PhysCellSpec> children;
+  // measured per-GPU HBM capacity in bytes (0 = unknown); leaf entries only
+  long long hbmBytes = 0;
+  // measured xGMI links (a, b, gbps, healthy); node-level entries only,
+  // emitted by rocm-topo-discover
+  std::vector<std::tuple<int, int, double, bool>> xgmiLinks;
 };
 struct VirtCellSpec {
   std::string typePath;  // "TOP.CHILD....", chain = first component
@@ -221,6 +245,9 @@ struct PodSpec {
   bool ignoreK8sSuggestedNodes = true;
   std::string groupName;
   std::map<int, int> groupPodNums;  // leafCellNum -> pod count
+  // optional: minimum measured HBM per leaf cell (bytes); leaves below are
+  // unavailable for this request (MI355X: a GPU reporting < 288 GB is sick)
+  long long hbmBytesPerCell = 0;
 };
 
 enum class Phase { Filtering, Preempting };
@@ -249,6 +276,7 @@ struct SchedulingRequest {
   int priority = 0;
   const std::set<std::string>* suggestedNodes = nullptr;
   bool ignoreSuggestedNodes = true;
+  long long hbmBytes = 0;  // minimum per-leaf HBM capacity (0 = any)
 };
 
 // ---------------------------------------------------------------------------
@@ -262,10 +290,12 @@ class TopoScheduler {
   TopoScheduler() = default;
   TopoScheduler(const ChainCellList& ccl, std::map<int, int> levelLeafNum, bool crossPriorityPack);
 
-  // Returns placement or empty with failedReason set.
+  // Returns placement or empty with failedReason set. minHbmBytes > 0
+  // filters out leaves whose measured HBM capacity falls short.
   bool Schedule(const std::map<int, int>& podLeafCellNums, int priority,
                 const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
-                Placement<Cell>* out, std::string* failedReason) const;
+                Placement<Cell>* out, std::string* failedReason,
+                long long minHbmBytes = 0) const;
 
  private:
   struct NodeView {
@@ -278,7 +308,8 @@ class TopoScheduler {
   };
   bool tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                              const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
-                             Placement<Cell>* out, std::string* failedReason) const;
+                             long long minHbmBytes, bool honorLinks, Placement<Cell>* out,
+                             std::string* failedReason) const;
 
   std::vector<Cell*> viewCells_;
   std::map<int, int> levelLeafNum_;
@@ -318,6 +349,13 @@ class HivedCore {
   // -- GPU/xGMI-level health (rocm-smi exporter / probe events): marks one
   // leaf cell; badness rolls up to pair/quad/node cells automatically --
   void setLeafCellHealthy(const std::string& node, int leafIndex, bool healthy);
+  // -- xGMI link health: a degraded link between two GPUs of one node marks
+  // the LINK (first-class), not the endpoint leaves: multi-GPU placements
+  // avoid co-placing the endpoints while 1-GPU work still uses them --
+  void setXgmiLinkHealthy(const std::string& node, int a, int b, bool healthy,
+                          double gbps = 0.0);
+  // per-node link table for inspect: (a, b, gbps, healthy)
+  std::vector<std::tuple<int, int, double, bool>> xgmiLinks(const std::string& node) const;
   std::vector<std::string> allNodes() const;
   std::set<std::string> badNodes() const { return badNodes_; }
 
@@ -454,7 +492,8 @@ class HivedCore {
                                      ChainCellList freeList, std::map<int, int> freeCellNum,
                                      const std::set<std::string>& suggestedNodes,
                                      bool ignoreSuggestedNodes,
-                                     std::unordered_map<VirtualCell*, PhysicalCell*>& bindings);
+                                     std::unordered_map<VirtualCell*, PhysicalCell*>& bindings,
+                                     long long minHbmBytes = 0, bool honorLinks = false);
 
  public:
   // state (public for inspect/bindings simplicity; external mutation forbidden)
@@ -478,6 +517,8 @@ class HivedCore {
   std::set<std::string> badNodes_;
   // leaves individually marked bad (GPU/xGMI level), independent of node health
   std::set<PhysicalCell*> badLeafMarks_;
+  // node -> (minIdx, maxIdx) -> link record (first-class xGMI link state)
+  std::map<std::string, std::map<std::pair<int, int>, XgmiLink>> xgmiLinks_;
   std::map<std::string, std::vector<std::string>> cellChains_;        // leaf type -> chains
   std::map<std::string, std::map<int, std::string>> cellTypes_;      // chain -> level -> type
   std::map<std::string, std::map<int, int>> leafCellNums_;           // chain -> level -> leaf num

@@ -65,6 +65,47 @@ void checkInvariants(const HivedCore& core) {
       }
     }
   }
+  // ---- xGMI link state: badLinksUnder roll-up + peer symmetry ----
+  {
+    // recompute badLinksUnder from the link registry and compare
+    std::unordered_map<const Cell*, int> expect;
+    for (auto& [node, links] : core.xgmiLinks_) {
+      (void)node;
+      for (auto& [key, l] : links) {
+        (void)key;
+        if (l.healthy) continue;
+        // peer symmetry on the endpoint leaves
+        auto hasPeer = [](PhysicalCell* from, PhysicalCell* to) {
+          return std::find(from->badLinkPeers.begin(), from->badLinkPeers.end(), to) !=
+                 from->badLinkPeers.end();
+        };
+        if (!hasPeer(l.a, l.b) || !hasPeer(l.b, l.a)) {
+          fail("bad xGMI link " + l.a->address + "<->" + l.b->address +
+               " not mirrored in badLinkPeers");
+        }
+        Cell* x = l.a;
+        Cell* y = l.b;
+        while (x != y && x != nullptr && y != nullptr) {
+          x = x->parent;
+          y = y->parent;
+        }
+        for (Cell* c = x; c != nullptr; c = c->parent) expect[c]++;
+      }
+    }
+    for (auto& [chain, ccl] : core.fullCellList_) {
+      (void)chain;
+      for (int l = ccl.top(); l >= kLowestLevel; l--) {
+        for (Cell* cc : ccl.at(l)) {
+          auto* c = static_cast<PhysicalCell*>(cc);
+          int want = expect.count(c) ? expect[c] : 0;
+          if (c->badLinksUnder != want) {
+            fail("badLinksUnder roll-up at " + c->address + ": " +
+                 std::to_string(c->badLinksUnder) + " != " + std::to_string(want));
+          }
+        }
+      }
+    }
+  }
   // ---- free list consistency + accounting ----
   for (auto& [chain, freeList] : core.freeCellList_) {
     for (int l = freeList.top(); l >= kLowestLevel; l--) {

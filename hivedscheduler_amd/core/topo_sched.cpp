@@ -7,6 +7,15 @@
 // an LCA-minimal, buddy-packed placement in O(depth * fanout) per pod — on the
 // fixed MI355X chain (leaf->pair->quad->node) this is a handful of comparisons,
 // which is what makes microsecond-scale Schedule() latency possible.
+//
+// MI355X-native additions over the reference:
+//  - per-leaf health (a sick GPU, not a whole node, is unavailable);
+//  - first-class xGMI link health: a degraded link between two GPUs excludes
+//    CO-PLACING its endpoints in one gang (the gang's all-reduce would ride
+//    the sick link) while both GPUs stay available for 1-GPU work — the
+//    reference's node/leaf health model cannot express this;
+//  - per-leaf HBM capacity: a request may demand hbmBytesPerCell; leaves
+//    whose measured capacity falls short are unavailable for it.
 #include <climits>
 
 #include "core.hpp"
@@ -37,25 +46,46 @@ TopoScheduler::TopoScheduler(const ChainCellList& ccl, std::map<int, int> levelL
 
 namespace {
 
-struct PickSession {
-  std::unordered_set<Cell*> taken;
-};
-
-// Leaf-granular health: a bad physical leaf (sick GPU / degraded xGMI link
-// endpoint) is never available; a virtual leaf bound to a bad physical leaf
-// (doomed-bad binding) is equally unavailable. This is finer than the
-// reference, whose health model stops at node granularity.
-bool leafHealthy(Cell* c) {
-  if (c->isPhysical()) return c->healthy;
-  PhysicalCell* p = static_cast<VirtualCell*>(c)->phys;
-  return p == nullptr || p->healthy;
+// Resolve a view cell (virtual or physical) to its physical identity, if any.
+// Virtual cells not (yet) bound have no physical identity: no link/HBM facts.
+PhysicalCell* physOf(Cell* c) {
+  if (c->isPhysical()) return static_cast<PhysicalCell*>(c);
+  return static_cast<VirtualCell*>(c)->phys;
 }
 
-// (availableTotal, availableFree): healthy leaves that are free or
-// lower-priority (preemptible), excluding ones already taken in this session.
-std::pair<int, int> availLeaves(Cell* c, int p, PickSession& s) {
+int badLinksUnderOf(Cell* c) {
+  PhysicalCell* p = physOf(c);
+  return p == nullptr ? 0 : p->badLinksUnder;
+}
+
+struct PickSession {
+  std::unordered_set<Cell*> taken;
+  // leaves excluded for THIS request because their bad-link peer may be
+  // co-placed (gang-wide: one session spans all pods of the gang)
+  std::unordered_set<Cell*> excluded;
+  // minimum per-leaf HBM capacity demanded by the request (0 = any)
+  long long minHbm = 0;
+};
+
+// Leaf-granular availability: a bad physical leaf (sick GPU) is never
+// available; a virtual leaf bound to a bad physical leaf (doomed-bad binding)
+// is equally unavailable; a leaf with measured HBM below the request's demand
+// is unavailable for this request. This is finer than the reference, whose
+// health model stops at node granularity.
+bool leafUsable(Cell* c, const PickSession& s) {
+  PhysicalCell* p = physOf(c);
+  if (p == nullptr) return true;  // unbound virtual: no physical facts yet
+  if (!p->healthy) return false;
+  if (s.minHbm > 0 && p->hbmBytes > 0 && p->hbmBytes < s.minHbm) return false;
+  return true;
+}
+
+// (availableTotal, availableFree): usable leaves that are free or
+// lower-priority (preemptible), excluding ones already taken or excluded in
+// this session.
+std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
   if (c->level == kLowestLevel) {
-    if (s.taken.count(c) || !leafHealthy(c)) return {0, 0};
+    if (s.taken.count(c) || s.excluded.count(c) || !leafUsable(c, s)) return {0, 0};
     if (c->priority == kFreePriority) return {1, 1};
     if (c->priority < p) return {1, 0};
     return {0, 0};
@@ -69,13 +99,111 @@ std::pair<int, int> availLeaves(Cell* c, int p, PickSession& s) {
   return {at, af};
 }
 
+void collectAvailableLeaves(Cell* c, int p, const PickSession& s, std::vector<Cell*>& out) {
+  if (c->level == kLowestLevel) {
+    if (!s.taken.count(c) && !s.excluded.count(c) && leafUsable(c, s) && c->priority < p) {
+      out.push_back(c);
+    }
+    return;
+  }
+  for (Cell* child : c->children) collectAvailableLeaves(child, p, s, out);
+}
+
+// Max independent set in the bad-link graph over `verts` (adjacency masks).
+// Bad links are rare (usually 0-2 per node), so the branch recursion is tiny;
+// callers cap the vertex count.
+unsigned long long maxIndepSet(const std::vector<unsigned long long>& adj,
+                               unsigned long long cand) {
+  if (cand == 0) return 0;
+  int v = __builtin_ctzll(cand);
+  unsigned long long bit = 1ull << v;
+  // exclude v
+  unsigned long long best = maxIndepSet(adj, cand & ~bit);
+  // include v (drop its bad-link neighbors)
+  unsigned long long with = bit | maxIndepSet(adj, cand & ~bit & ~adj[v]);
+  if (__builtin_popcountll(with) > __builtin_popcountll(best)) best = with;
+  return best;
+}
+
+// Link-clean analysis of `viewCell` for a request needing `need` leaves that
+// will all communicate (one gang on one node): find the largest set of
+// available leaves with no degraded xGMI link INSIDE the set. Returns the
+// max clean count; when `excludeOut` is non-null and the clean capacity
+// covers `need`, fills it with the leaves to exclude (the bad-link-incident
+// leaves NOT in the chosen independent set).
+int cleanAvailAnalysis(Cell* viewCell, int p, const PickSession& s, int need,
+                       std::vector<Cell*>* excludeOut) {
+  auto [availTotal, availFree] = availLeaves(viewCell, p, s);
+  (void)availFree;
+  if (badLinksUnderOf(viewCell) == 0) return availTotal;
+  std::vector<Cell*> leaves;
+  collectAvailableLeaves(viewCell, p, s, leaves);
+  // map physical leaf -> view leaf to resolve link endpoints in this view
+  std::unordered_map<PhysicalCell*, int> physIdx;
+  for (size_t i = 0; i < leaves.size(); i++) {
+    PhysicalCell* ph = physOf(leaves[i]);
+    if (ph != nullptr) physIdx[ph] = static_cast<int>(i);
+  }
+  // vertices = available leaves incident to a bad link whose peer is also
+  // available in this view
+  std::vector<int> verts;
+  std::vector<unsigned long long> adj;
+  std::unordered_map<int, int> leafToVert;
+  auto vertOf = [&](int leafIdx) {
+    auto it = leafToVert.find(leafIdx);
+    if (it != leafToVert.end()) return it->second;
+    int v = static_cast<int>(verts.size());
+    leafToVert[leafIdx] = v;
+    verts.push_back(leafIdx);
+    adj.push_back(0);
+    return v;
+  };
+  for (size_t i = 0; i < leaves.size(); i++) {
+    PhysicalCell* ph = physOf(leaves[i]);
+    if (ph == nullptr || ph->badLinkPeers.empty()) continue;
+    for (PhysicalCell* peer : ph->badLinkPeers) {
+      auto it = physIdx.find(peer);
+      if (it == physIdx.end()) continue;
+      int a = vertOf(static_cast<int>(i));
+      int b = vertOf(it->second);
+      if (a < 64 && b < 64) {
+        adj[a] |= 1ull << b;
+        adj[b] |= 1ull << a;
+      }
+    }
+  }
+  int nv = static_cast<int>(verts.size());
+  if (nv == 0) return availTotal;
+  if (nv > 60) return availTotal - nv;  // degenerate; be conservative
+  unsigned long long all = nv == 64 ? ~0ull : ((1ull << nv) - 1);
+  unsigned long long chosen = maxIndepSet(adj, all);
+  int clean = availTotal - nv + __builtin_popcountll(chosen);
+  if (excludeOut != nullptr && clean >= need) {
+    for (int v = 0; v < nv; v++) {
+      if (!(chosen & (1ull << v))) excludeOut->push_back(leaves[verts[v]]);
+    }
+  }
+  return clean;
+}
+
+// Exclude, for the rest of this session (= this gang), the bad-link-incident
+// leaves that a clean `need`-leaf placement on `viewCell` must avoid. No-op
+// when clean capacity cannot cover the need (a dirty placement is then
+// allowed: capacity guarantees outrank link quality).
+void applyLinkExclusions(Cell* viewCell, int need, int p, PickSession& s) {
+  if (need < 2 || badLinksUnderOf(viewCell) == 0) return;
+  std::vector<Cell*> toExclude;
+  cleanAvailAnalysis(viewCell, p, s, need, &toExclude);
+  for (Cell* c : toExclude) s.excluded.insert(c);
+}
+
 // Lowest LCA level achievable for q leaves anywhere inside `cell`
 // (INT_MAX if the subtree cannot hold q). This is the lookahead that makes
 // the descent LCA-minimal: a child with MORE availability may reach a LOWER
 // LCA (e.g. free = {0,1,2} in quad A vs {5,6} straddling quad B's pairs —
 // A serves a pair request at pair level, B only at quad level). Verified
 // against brute force in tests/test_placement_optimality.py.
-int bestLCALevel(Cell* c, int q, int p, PickSession& s) {
+int bestLCALevel(Cell* c, int q, int p, const PickSession& s) {
   auto [t, f] = availLeaves(c, p, s);
   (void)f;
   if (t < q) return INT_MAX;
@@ -101,8 +229,11 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
   for (int i = 0; i < n; i++) av[i] = availLeaves(cell->children[i], p, s);
 
   // A single child can hold the whole request: descend into the one that
-  // reaches the lowest LCA, then needs the fewest preemptions, then is the
-  // tightest fit (packing).
+  // reaches the lowest LCA, then needs the fewest preemptions, then — for
+  // 1-leaf requests — prefers link-degraded subtrees (parking 1-GPU work on
+  // degraded pairs keeps clean pairs free for multi-GPU gangs; multi-leaf
+  // requests avoid degraded endpoints via session exclusions instead), then
+  // is the tightest fit (packing).
   int best = -1, bestLca = INT_MAX;
   for (int i = 0; i < n; i++) {
     if (av[i].first >= q) {
@@ -114,9 +245,18 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
       }
       int needPreempt = std::max(0, q - av[i].second);
       int bestPreempt = std::max(0, q - av[best].second);
+      int linkKey = badLinksUnderOf(cell->children[i]);
+      int bestLinkKey = badLinksUnderOf(cell->children[best]);
+      if (q == 1) {
+        linkKey = -linkKey;
+        bestLinkKey = -bestLinkKey;
+      }
       if (lca < bestLca ||
-          (lca == bestLca && (needPreempt < bestPreempt ||
-                              (needPreempt == bestPreempt && av[i].first < av[best].first)))) {
+          (lca == bestLca &&
+           (needPreempt < bestPreempt ||
+            (needPreempt == bestPreempt &&
+             (linkKey < bestLinkKey ||
+              (linkKey == bestLinkKey && av[i].first < av[best].first)))))) {
         best = i;
         bestLca = lca;
       }
@@ -146,12 +286,7 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
 
 std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>& suggestedNodes,
                                            bool ignoreSuggestedNodes) {
-  PhysicalCell* pc = nullptr;
-  if (c->isPhysical()) {
-    pc = static_cast<PhysicalCell*>(c);
-  } else {
-    pc = static_cast<VirtualCell*>(c)->phys;
-  }
+  PhysicalCell* pc = physOf(c);
   if (pc == nullptr) return {true, true};
   bool suggested =
       ignoreSuggestedNodes || (!pc->nodes.empty() && suggestedNodes.count(pc->nodes[0]) > 0);
@@ -162,13 +297,15 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 
 bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                                           const std::set<std::string>& suggestedNodes,
-                                          bool ignoreSuggestedNodes, Placement<Cell>* out,
+                                          bool ignoreSuggestedNodes, long long minHbmBytes,
+                                          bool honorLinks, Placement<Cell>* out,
                                           std::string* failedReason) const {
   // Build and sort the cluster view: healthy > suggested > same-priority used
   // (desc, packing) > higher-priority used (asc, stay away).
   std::vector<NodeView> cv;
   cv.reserve(viewCells_.size());
   PickSession probe;  // empty: availability before any placement
+  probe.minHbm = minHbmBytes;
   for (Cell* c : viewCells_) {
     NodeView n;
     n.c = c;
@@ -182,8 +319,14 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
       }
     }
     // health-aware availability: bad leaves never count, so partially-bad
-    // nodes stay usable for their healthy pairs/quads
-    n.freeAtPriority = availLeaves(c, priority, probe).first;
+    // nodes stay usable for their healthy pairs/quads. In link-honoring mode
+    // a node's capacity is its CLEAN capacity (largest co-placeable set with
+    // no degraded xGMI link inside); the dirty retry lifts that.
+    if (honorLinks && badLinksUnderOf(c) > 0) {
+      n.freeAtPriority = cleanAvailAnalysis(c, priority, probe, INT_MAX, nullptr);
+    } else {
+      n.freeAtPriority = availLeaves(c, priority, probe).first;
+    }
     auto [healthy, suggested] = healthyAndSuggested(c, suggestedNodes, ignoreSuggestedNodes);
     n.healthy = healthy;
     n.suggested = suggested;
@@ -248,6 +391,19 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
 
   // Pick leaf cells inside each pod's node.
   PickSession session;
+  session.minHbm = minHbmBytes;
+  if (honorLinks) {
+    // Gang-wide link cleanliness per node: every pod of the gang on one node
+    // communicates with every other (the gang's collective), so exclusions
+    // are computed from the node's TOTAL gang demand before any pick.
+    std::map<int, int> nodeDemand;  // nodeIndex -> total gang leaves
+    for (size_t i = 0; i < sortedLeafNums.size(); i++) {
+      nodeDemand[pickedNodeIndices[i]] += sortedLeafNums[i];
+    }
+    for (auto& [nodeIndex, demand] : nodeDemand) {
+      if (demand >= 2) applyLinkExclusions(cv[nodeIndex].c, demand, priority, session);
+    }
+  }
   out->clear();
   for (size_t i = 0; i < sortedLeafNums.size(); i++) {
     int q = sortedLeafNums[i];
@@ -263,20 +419,31 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
 bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int priority,
                              const std::set<std::string>& suggestedNodes,
                              bool ignoreSuggestedNodes, Placement<Cell>* out,
-                             std::string* failedReason) const {
+                             std::string* failedReason, long long minHbmBytes) const {
   std::vector<int> sortedLeafNums;
   for (auto& [leafNum, podNum] : podLeafCellNums) {
     for (int i = 0; i < podNum; i++) sortedLeafNums.push_back(leafNum);
   }
-  // First try without preemption (only free cells), then allow preempting
-  // lower priorities.
-  if (tryScheduleAtPriority(sortedLeafNums, kOpportunisticPriority, suggestedNodes,
-                            ignoreSuggestedNodes, out, failedReason)) {
-    return true;
+  bool anyBadLinks = false;
+  for (Cell* c : viewCells_) {
+    if (badLinksUnderOf(c) > 0) {
+      anyBadLinks = true;
+      break;
+    }
   }
+  // Attempt ladder: free cells with clean links, free cells ignoring link
+  // state (capacity outranks link quality), then the same two with
+  // preemption. Avoiding a preemption (killing pods) outranks avoiding a
+  // degraded link (slow xGMI), hence free+dirty before preempt+clean.
+  auto attempt = [&](int p, bool honorLinks) {
+    return tryScheduleAtPriority(sortedLeafNums, p, suggestedNodes, ignoreSuggestedNodes,
+                                 minHbmBytes, honorLinks, out, failedReason);
+  };
+  if (attempt(kOpportunisticPriority, true)) return true;
+  if (anyBadLinks && attempt(kOpportunisticPriority, false)) return true;
   if (priority > kOpportunisticPriority) {
-    return tryScheduleAtPriority(sortedLeafNums, priority, suggestedNodes, ignoreSuggestedNodes,
-                                 out, failedReason);
+    if (attempt(priority, true)) return true;
+    if (anyBadLinks && attempt(priority, false)) return true;
   }
   return false;
 }
@@ -294,7 +461,7 @@ bool IntraVCScheduler::schedule(const SchedulingRequest& sr, Placement<VirtualCe
   Placement<Cell> generic;
   if (scheduler == nullptr ||
       !scheduler->Schedule(sr.podLeafCellNums, sr.priority, *sr.suggestedNodes,
-                           sr.ignoreSuggestedNodes, &generic, failedReason)) {
+                           sr.ignoreSuggestedNodes, &generic, failedReason, sr.hbmBytes)) {
     if (failedReason->empty()) *failedReason = "no scheduler for request";
     *failedReason += " when scheduling in VC " + sr.vc;
     return false;

@@ -135,7 +135,10 @@ class HivedScheduler:
 
     def on_health_report(self, node: str, report: dict) -> dict:
         """GPU-level health from a node agent: marks individual leaf cells
-        bad/healthy (xGMI-degraded pairs arrive as bad endpoint leaves)."""
+        bad/healthy, and applies xGMI LINK state first-class — a degraded
+        link ("links": [{"a", "b", "healthy", "gbps"}]) makes multi-GPU
+        placements avoid co-placing its endpoints while both GPUs stay
+        schedulable for 1-GPU work."""
         with self.lock:
             self.health_reports[node] = report
             applied = {}
@@ -147,21 +150,47 @@ class HivedScheduler:
                 healthy = bool(gpu.get("healthy", True))
                 self.algorithm.set_leaf_cell_healthy(node, idx, healthy)
                 applied[idx_str] = healthy
-            return {"node": node, "applied": applied}
+            applied_links = {}
+            for link in (report.get("links") or []):
+                try:
+                    a, b = int(link["a"]), int(link["b"])
+                except (KeyError, TypeError, ValueError):
+                    continue
+                healthy = bool(link.get("healthy", True))
+                gbps = float(link.get("gbps") or 0.0)
+                self.algorithm.set_xgmi_link_healthy(node, a, b, healthy, gbps)
+                applied_links[f"{min(a, b)}-{max(a, b)}"] = healthy
+            out = {"node": node, "applied": applied}
+            if applied_links:
+                out["appliedLinks"] = applied_links
+            return out
+
+    def get_xgmi_links(self, node: str) -> list:
+        with self.lock:
+            return self.algorithm.get_xgmi_links(node)
 
     def get_health_reports(self) -> Dict[str, dict]:
         with self.lock:
             return dict(self.health_reports)
 
     def on_probe_result(self, result: dict) -> dict:
-        """Placement-probe result from a node agent; an unhealthy verdict
-        (busbw below the xGMI floor) marks the probed leaf cells bad."""
+        """Placement-probe result from a node agent. An unhealthy 2-GPU
+        verdict (busbw below the xGMI floor) localizes to ONE link and marks
+        it first-class — the endpoint GPUs stay schedulable for 1-GPU work.
+        Larger probes cannot localize the sick link from busbw alone, so
+        their leaves are marked bad pending the agent's p2p matrix sweep."""
         verdict = self.probe_manager.report(result)
         if not verdict["healthy"]:
             node = result.get("node", "")
+            indices = [int(i) for i in result.get("leafCellIndices", [])]
             with self.lock:
-                for idx in result.get("leafCellIndices", []):
-                    self.algorithm.set_leaf_cell_healthy(node, int(idx), False)
+                if len(indices) == 2:
+                    busbw = float(result.get("busbw_gbps") or 0.0)
+                    self.algorithm.set_xgmi_link_healthy(node, indices[0], indices[1],
+                                                         False, busbw)
+                else:
+                    for idx in indices:
+                        self.algorithm.set_leaf_cell_healthy(node, idx, False)
         return verdict
 
     def _add_bound_pod(self, pod: dict) -> None:
@@ -274,12 +303,18 @@ class HivedScheduler:
                 _DECISIONS.labels("wait").inc()
             self.pod_statuses[uid] = PodScheduleStatus(
                 pod=pod, state=POD_WAITING, pod_scheduling_spec=spec)
-            if self.waiting_block_ms > 0:
-                time.sleep(self.waiting_block_ms / 1e3)
             reason = "Pod is waiting for preemptible or free resource to appear"
             if result.wait_reason:
                 reason += ": " + result.wait_reason
-            return {"FailedNodes": {constants.ComponentName: reason}}
+            wait_response = {"FailedNodes": {constants.ComponentName: reason}}
+        # FIFO throughput block (waitingPodSchedulingBlockMilliSec) runs
+        # OUTSIDE the scheduler lock: sleeping under it would stall every
+        # concurrent filter/bind/health call for the block duration (the
+        # reference's Go server sleeps on its own goroutine; scheduler.go:
+        # 560-586 holds no lock during the block either)
+        if self.waiting_block_ms > 0:
+            time.sleep(self.waiting_block_ms / 1e3)
+        return wait_response
 
     def bind(self, args: Dict[str, Any]) -> Dict[str, Any]:
         t0 = time.perf_counter()

@@ -79,6 +79,7 @@ class SimScheduler:
         pinned_cell_id: str = "",
         lazy_preemption: bool = False,
         ignore_suggested: bool = True,
+        hbm_bytes_per_cell: int = 0,
     ) -> PodSchedulingSpec:
         from ..api.types import AffinityGroupMemberSpec, AffinityGroupSpec
 
@@ -90,6 +91,7 @@ class SimScheduler:
             leafCellNumber=leaf_cells,
             lazyPreemptionEnable=lazy_preemption,
             ignoreK8sSuggestedNodes=ignore_suggested,
+            hbmBytesPerCell=hbm_bytes_per_cell,
         )
         if group is not None:
             spec.affinityGroup = AffinityGroupSpec(

@@ -16,12 +16,19 @@ from typing import Any, Dict, Optional
 
 from fastapi import FastAPI, Request, Response
 from fastapi.responses import JSONResponse, PlainTextResponse
+from starlette.concurrency import run_in_threadpool
 
 from ..api import constants
 from ..api.types import WebServerError
 from ..scheduler import HivedScheduler
 
 log = logging.getLogger("hivedscheduler.webserver")
+
+# Scheduler calls run in the threadpool, never on the event loop: they take
+# the scheduler lock, and a waiting pod's FIFO block (or a slow filter) would
+# otherwise stall every concurrent request — /healthz, health intake, probe
+# polling. The reference's Go server serves each request on its own goroutine
+# (webserver.go:93-155); run_in_threadpool is the asyncio equivalent.
 
 
 def create_app(scheduler: HivedScheduler) -> FastAPI:
@@ -61,7 +68,7 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
         args = await _json_body(request)
         if not isinstance(args.get("Pod"), dict):
             raise WebServerError.bad_request("ExtenderArgs.Pod is missing")
-        return JSONResponse(scheduler.filter(args))
+        return JSONResponse(await run_in_threadpool(scheduler.filter, args))
 
     @app.post(constants.BindPath)
     async def bind_verb(request: Request):
@@ -70,7 +77,7 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
             if not args.get(field):
                 raise WebServerError.bad_request(f"ExtenderBindingArgs.{field} is missing")
         try:
-            return scheduler.bind(args)
+            return await run_in_threadpool(scheduler.bind, args)
         except WebServerError as e:
             # binding errors are returned in-band so the default scheduler
             # surfaces them on the pod (reference webserver.go:194-215)
@@ -81,54 +88,59 @@ def create_app(scheduler: HivedScheduler) -> FastAPI:
         args = await _json_body(request)
         if not isinstance(args.get("Pod"), dict):
             raise WebServerError.bad_request("ExtenderPreemptionArgs.Pod is missing")
-        return JSONResponse(scheduler.preempt(args))
+        return JSONResponse(await run_in_threadpool(scheduler.preempt, args))
 
     # ---- inspect API ----
     @app.get(constants.AffinityGroupsPath)
     async def affinity_groups():
-        return scheduler.get_all_affinity_groups()
+        return await run_in_threadpool(scheduler.get_all_affinity_groups)
 
     @app.get(constants.AffinityGroupsPath + "{name:path}")
     async def affinity_group(name: str):
-        return scheduler.get_affinity_group(name)
+        return await run_in_threadpool(scheduler.get_affinity_group, name)
 
     @app.get(constants.ClusterStatusPath)
     async def cluster_status():
-        return scheduler.get_cluster_status()
+        return await run_in_threadpool(scheduler.get_cluster_status)
 
     @app.get(constants.PhysicalClusterPath)
     async def physical_cluster():
-        return scheduler.get_physical_cluster_status()
+        return await run_in_threadpool(scheduler.get_physical_cluster_status)
 
     @app.get(constants.VirtualClustersPath)
     async def virtual_clusters():
-        return scheduler.get_all_virtual_clusters_status()
+        return await run_in_threadpool(scheduler.get_all_virtual_clusters_status)
 
     @app.get(constants.VirtualClustersPath + "{vc}")
     async def virtual_cluster(vc: str):
-        return scheduler.get_virtual_cluster_status(vc)
+        return await run_in_threadpool(scheduler.get_virtual_cluster_status, vc)
 
     # ---- GPU-level health intake (node agents) ----
     @app.post("/v1/health/nodes/{node}")
     async def health_report(node: str, report: Dict[str, Any]):
-        return scheduler.on_health_report(node, report)
+        return await run_in_threadpool(scheduler.on_health_report, node, report)
 
     @app.get("/v1/inspect/health")
     async def health_reports():
-        return scheduler.get_health_reports()
+        return await run_in_threadpool(scheduler.get_health_reports)
+
+    @app.get("/v1/inspect/links/{node}")
+    async def xgmi_links(node: str):
+        # first-class xGMI link state: [{a, b, gbps, healthy}, ...]
+        return await run_in_threadpool(scheduler.get_xgmi_links, node)
 
     # ---- post-bind placement probes (agents poll tasks, post results) ----
     @app.get("/v1/health/probes/{node}")
     async def probe_tasks(node: str):
-        return scheduler.probe_manager.poll(node)
+        return await run_in_threadpool(scheduler.probe_manager.poll, node)
 
     @app.post("/v1/health/probes")
     async def probe_result(result: Dict[str, Any]):
-        return scheduler.on_probe_result(result)
+        return await run_in_threadpool(scheduler.on_probe_result, result)
 
     @app.get("/v1/inspect/probes/{group:path}")
     async def probe_results(group: str):
-        return scheduler.probe_manager.group_results(group)
+        return await run_in_threadpool(scheduler.probe_manager.group_results, group)
 
     # ---- metrics ----
     @app.get(constants.MetricsPath)

@@ -163,11 +163,18 @@ def test_fuzz_extended(seed):
                         del preempting[key]
                     else:
                         del preempting[key]
-            elif op < 0.83:
+            elif op < 0.80:
                 sim.alg.update_node(rng.choice(nodes), rng.random() < 0.7)
-            elif op < 0.97:
+            elif op < 0.90:
                 sim.alg.set_leaf_cell_healthy(rng.choice(nodes), rng.randrange(8),
                                               rng.random() < 0.7)
+            elif op < 0.97:
+                # xGMI link flapping: first-class link marks compose with
+                # leaf/node health and placements must stay invariant-clean
+                a, b = rng.sample(range(8), 2)
+                sim.alg.set_xgmi_link_healthy(rng.choice(nodes), a, b,
+                                              rng.random() < 0.6,
+                                              rng.uniform(5.0, 160.0))
             else:
                 # crash-recovery: fresh algorithm, replay live pods from bind
                 # infos (pods are the database; Preempting state is volatile)

@@ -137,3 +137,95 @@ def test_probe_degraded_link_marks_cells_bad(client):
     assert rr.json().get("NodeNames")
     st = client.get(constants.AffinityGroupsPath + "ns/p2").json()
     assert set(st["physicalPlacement"]["node1"]).isdisjoint({0, 1})
+
+
+def test_health_intake_applies_links(client):
+    """A degraded-link report marks the LINK first-class: the endpoint GPUs
+    stay schedulable (cells stay Healthy) while the link shows in inspect."""
+    r = client.post("/v1/health/nodes/node1", json={
+        "gpus": {}, "links": [{"a": 0, "b": 1, "healthy": False, "gbps": 12.5}]})
+    assert r.status_code == 200
+    assert r.json()["appliedLinks"] == {"0-1": False}
+    links = client.get("/v1/inspect/links/node1").json()
+    assert links == [{"a": 0, "b": 1, "gbps": 12.5, "healthy": False}]
+    # no cell went Bad: link degradation is not leaf badness
+    status = client.get(constants.PhysicalClusterPath).json()
+
+    def walk(c):
+        yield c
+        for ch in c.get("cellChildren") or []:
+            yield from walk(ch)
+
+    cells = [c for top in status for c in walk(top)]
+    assert all(c.get("cellHealthiness") == "Healthy" for c in cells)
+    assert any(c.get("badXgmiLinksUnder") for c in cells)
+    # a 2-GPU filter avoids the degraded pair
+    pod = make_pod("p-l1", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 2})
+    fr = client.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+    assert fr.status_code == 200 and fr.json()["NodeNames"] == ["node1"]
+
+
+def test_probe_result_pair_marks_link_not_leaves(client):
+    """A failed 2-GPU placement probe localizes to ONE link; the endpoint
+    GPUs stay schedulable for 1-GPU work."""
+    r = client.post("/v1/health/probes", json={
+        "ok": True, "healthy": False, "node": "node1", "group": "g1",
+        "leafCellIndices": [2, 3], "busbw_gbps": 8.0})
+    assert r.status_code == 200
+    links = {(l["a"], l["b"]): l for l in client.get("/v1/inspect/links/node1").json()}
+    assert not links[(2, 3)]["healthy"]
+    assert links[(2, 3)]["gbps"] == 8.0
+    status = client.get(constants.PhysicalClusterPath).json()
+
+    def walk(c):
+        yield c
+        for ch in c.get("cellChildren") or []:
+            yield from walk(ch)
+
+    leaves = [c for top in status for c in walk(top) if c.get("leafCellIndex") is not None]
+    assert all(c["cellHealthiness"] == "Healthy" for c in leaves)
+
+
+def test_event_loop_not_blocked_by_fifo_wait():
+    """A waiting pod's FIFO block (50 ms) must not stall concurrent requests:
+    scheduler calls run in the threadpool and the sleep happens outside the
+    scheduler lock (reference: goroutine-per-request)."""
+    import threading
+    import time as _time
+
+    from hivedscheduler_amd.api.config import Config as _C  # noqa: F401
+
+    cfg = mi355x_cluster_config(num_nodes=1)
+    cfg.waitingPodSchedulingBlockMilliSec = 200
+    sched = HivedScheduler(cfg)
+    sched.on_node_add(make_node("node1"))
+    from starlette.testclient import TestClient
+
+    c = TestClient(create_app(sched), raise_server_exceptions=False)
+    # an unsatisfiable request -> wait path with the 200 ms FIFO block
+    pod = make_pod("p-wait", {"virtualCluster": "VC1", "priority": 0, "leafCellNumber": 8,
+                              "affinityGroup": {"name": "gw", "members": [
+                                  {"podNumber": 2, "leafCellNumber": 8}]}})
+    timings = {}
+
+    def waiter():
+        t0 = _time.perf_counter()
+        c.post(constants.FilterPath, json={"Pod": pod, "NodeNames": ["node1"]})
+        timings["wait"] = _time.perf_counter() - t0
+
+    th = threading.Thread(target=waiter)
+    th.start()
+    _time.sleep(0.05)  # let the filter reach its FIFO sleep
+    t0 = _time.perf_counter()
+    r = c.get("/healthz")
+    healthz_t = _time.perf_counter() - t0
+    assert r.status_code == 200
+    t0 = _time.perf_counter()
+    r2 = c.post("/v1/health/nodes/node1", json={"gpus": {"0": {"healthy": True}}})
+    health_post_t = _time.perf_counter() - t0
+    assert r2.status_code == 200
+    th.join(timeout=5)
+    assert timings["wait"] >= 0.2, f"FIFO block missing: {timings}"
+    # concurrent requests must not have waited for the 200 ms block
+    assert healthz_t < 0.15, f"/healthz stalled {healthz_t * 1e3:.0f} ms behind FIFO block"
+    assert health_post_t < 0.15, f"health POST stalled {health_post_t * 1e3:.0f} ms"
