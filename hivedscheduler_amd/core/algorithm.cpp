@@ -202,10 +202,13 @@ ScheduleResult HivedCore::generateResult(const Placement<PhysicalCell>& phys, bo
     return r;
   }
   if (!victims.empty()) {
-    // K8s preempts one node per round: report victims of one node
-    // (deterministic first node; randomness is not needed for completeness).
+    // K8s preempts one node per round: report victims of ONE node, chosen
+    // uniformly at random (seeded) so victim churn spreads across nodes
+    // under contention instead of herding on the first map key (reference
+    // utils.go:82-103 randomizes for the same reason).
     r.kind = ScheduleResult::Kind::Preempt;
     auto it = victims.begin();
+    std::advance(it, static_cast<long>(victimRng_() % victims.size()));
     r.victimNode = it->first;
     r.victimPodKeys.assign(it->second.begin(), it->second.end());
     return r;

@@ -22,6 +22,7 @@
 #include <map>
 #include <memory>
 #include <optional>
+#include <random>
 #include <set>
 #include <stdexcept>
 #include <string>
@@ -525,6 +526,9 @@ class HivedCore {
   // pinned: vc -> pinnedId -> physical cell
   std::map<std::string, std::map<std::string, PhysicalCell*>> pinnedPhysical_;
 
+  // seeded PRNG for victim-node selection (deterministic for tests/fuzz,
+  // spreads victim churn across nodes like the reference's rand)
+  std::minstd_rand victimRng_{12345};
   // cell ownership
   std::vector<std::unique_ptr<Cell>> cellStore_;
   // node name -> leaf cells on that node (for health propagation)

@@ -163,6 +163,14 @@ def test_rccl_cell_probe_binary():
 
 @needs_gpu
 def test_rocm_topo_discover():
+    """Discovery emits measured facts (VRAM, link table) and the output is
+    consumed end-to-end: YAML -> Config -> live core with link state."""
+    from hivedscheduler_amd.sim import SimScheduler
+    from hivedscheduler_amd.topo.discover import (
+        cluster_config_from_discovery,
+        parse_discovery_output,
+    )
+
     binary = os.path.join(REPO, "native", "rocm-topo-discover")
     assert os.path.exists(binary)
     out = subprocess.run([binary, "--node-name", "testnode"], capture_output=True, text=True,
@@ -170,6 +178,20 @@ def test_rocm_topo_discover():
     assert out.returncode == 0, out.stderr
     assert "nodeName: testnode" in out.stdout
     assert "physicalCells:" in out.stdout
+    doc = parse_discovery_output(out.stdout)
+    n = int(doc["numGpus"])
+    assert doc["gpus"][0]["vramBytes"] > 200 * 1024 ** 3, "measured VRAM missing"
+    if n >= 2:
+        # the measured per-link table is present and parses into core state
+        links_yaml = doc["physicalCells"][0].get("xgmiLinks") or []
+        assert links_yaml, "xgmiLinks table missing from discovery output"
+        if n == 8:
+            cfg = cluster_config_from_discovery([doc])
+            sim = SimScheduler(cfg)
+            links = sim.alg.get_xgmi_links("testnode")
+            assert len(links) == 28, f"8-GPU full mesh should have 28 links: {len(links)}"
+            assert all(l["healthy"] for l in links)
+            sim.alg._core.check_invariants()
 
 
 @needs_gpu

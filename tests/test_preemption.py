@@ -167,3 +167,24 @@ def test_lazy_preemption_downgrade():
     else:
         assert r.kind == "bind"
         assert "lazy" in {g["name"] for g in sim.alg.get_all_affinity_groups()}
+
+
+def test_victim_node_spread():
+    """Victim-node selection is randomized (seeded): repeated filter-phase
+    preemption probes for a cross-node gang report victims from DIFFERENT
+    nodes over rounds, spreading churn instead of herding on one node
+    (reference utils.go:82-103 behavior)."""
+    sim = SimScheduler(mi355x_cluster_config(
+        num_nodes=2, vcs={"VC1": [("MI355X-NODE", 2)]}))
+    for i in range(2):
+        r = sim.schedule(f"ns/op{i}", sim.pod_spec(leaf_cells=8, priority=-1))
+        assert r.kind == "bind", i
+    spec = sim.pod_spec(leaf_cells=8, group="gang", members=[(2, 8)])
+    seen = set()
+    for _ in range(30):
+        r = sim.schedule("ns/gang-0", spec, commit=False)
+        assert r.kind == "preempt"
+        seen.add(r.victim_node)
+        if len(seen) == 2:
+            break
+    assert seen == {"node1", "node2"}, f"victim churn herded on {seen}"
