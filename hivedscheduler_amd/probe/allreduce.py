@@ -88,19 +88,27 @@ class CellProbeRunner:
         return os.path.exists(self.binary) and os.access(self.binary, os.X_OK)
 
     def probe_cell(self, leaf_indices: List[int], size_mb: int = 64, iters: int = 20,
-                   timeout_s: float = 300.0) -> dict:
+                   timeout_s: float = 300.0, p2p_matrix: bool = False) -> dict:
         """All-reduce over the given GPU indices (one node). Returns the
         parsed probe JSON plus a health verdict. A hung probe (RCCL init can
         take minutes on a cold box; a truly wedged link hangs forever) is
         killed at timeout_s and reported as not-ok rather than raising — the
-        agent loop must survive it."""
+        agent loop must survive it.
+
+        p2p_matrix=True also measures every pair's p2p copy bandwidth
+        (worse direction), localizing a low collective busbw to specific
+        links: the result gains "p2p_matrix" ({"i-j": gbps} in VISIBLE-device
+        ordinals) and "suspect_links" ([[leafA, leafB, gbps], ...] in the
+        caller's leaf indices) for pairs below min_busbw_gbps.
+        """
         env = dict(os.environ)
         env["HIP_VISIBLE_DEVICES"] = ",".join(str(i) for i in leaf_indices)
+        cmd = [self.binary, "--size-mb", str(size_mb), "--iters", str(iters)]
+        if p2p_matrix and len(leaf_indices) > 1:
+            cmd.append("--p2p-matrix")
         try:
-            out = subprocess.run(
-                [self.binary, "--size-mb", str(size_mb), "--iters", str(iters)],
-                env=env, capture_output=True, text=True, timeout=timeout_s,
-            )
+            out = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                                 timeout=timeout_s)
         except subprocess.TimeoutExpired:
             return {"ok": False, "error": f"probe timed out after {timeout_s:.0f}s",
                     "timeout": True, "leaf_indices": leaf_indices}
@@ -111,4 +119,13 @@ class CellProbeRunner:
         result["ok"] = True
         n = len(leaf_indices)
         result["healthy"] = bool(n <= 1 or result.get("busbw_gbps", 0.0) >= self.min_busbw_gbps)
+        if result.get("p2p_matrix"):
+            suspects = []
+            for key, gbps in result["p2p_matrix"].items():
+                if gbps < self.min_busbw_gbps:
+                    vi, vj = (int(x) for x in key.split("-"))
+                    suspects.append([leaf_indices[vi], leaf_indices[vj], gbps])
+            result["suspect_links"] = suspects
+            if suspects:
+                result["healthy"] = False
         return result

@@ -183,8 +183,14 @@ class HivedScheduler:
         if not verdict["healthy"]:
             node = result.get("node", "")
             indices = [int(i) for i in result.get("leafCellIndices", [])]
+            suspects = result.get("suspect_links") or []
             with self.lock:
-                if len(indices) == 2:
+                if suspects:
+                    # the probe's p2p matrix localized the sick link(s)
+                    for a, b, gbps in suspects:
+                        self.algorithm.set_xgmi_link_healthy(node, int(a), int(b),
+                                                             False, float(gbps))
+                elif len(indices) == 2:
                     busbw = float(result.get("busbw_gbps") or 0.0)
                     self.algorithm.set_xgmi_link_healthy(node, indices[0], indices[1],
                                                          False, busbw)

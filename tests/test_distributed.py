@@ -11,7 +11,7 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def run_torchrun(script_args, nproc=2, timeout=300):
+def run_torchrun(script_args, nproc=2, timeout=300, port=29517):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
     # force the gloo/CPU path even on a GPU box: these tests model world>1
@@ -21,7 +21,7 @@ def run_torchrun(script_args, nproc=2, timeout=300):
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", f"--nproc-per-node={nproc}",
-        "--master-addr", "127.0.0.1", "--master-port", "29517",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
     ] + script_args
     return subprocess.run(cmd, capture_output=True, text=True, timeout=timeout, env=env,
                           cwd=REPO)
@@ -95,5 +95,38 @@ dist.destroy_process_group()
     assert out.returncode == 0, out.stderr[-3000:]
     line = [l for l in out.stdout.splitlines() if l.startswith("SUBGROUP_RESULT")][0]
     res = json.loads(line.split(" ", 1)[1])
-    assert set(res) == {"1", "2"}, res
-    assert all(v > 0 for v in res.values()), res
+    # 1-GPU entry is the device-local copy, reported separately
+    assert res["hbm_copy_gbps"] > 0, res
+    assert set(res["busbw"]) == {"2"}, res
+    assert all(v > 0 for v in res["busbw"].values()), res
+
+
+SUBGROUP_SCRIPT = """
+import os, sys, json
+sys.path.insert(0, %r)
+import torch.distributed as dist
+dist.init_process_group(backend="gloo")
+import bench
+world = dist.get_world_size()
+rank = dist.get_rank()
+res = bench.rccl_cell_probes(world, rank, rank)
+if rank == 0:
+    print("SUBGROUP_RESULT " + json.dumps(res))
+dist.destroy_process_group()
+""" % REPO
+
+
+@pytest.mark.parametrize("world,expected", [(4, {"2", "4"}), (8, {"2", "4", "8"})])
+def test_cell_probe_subgroups_gloo_world_4_8(tmp_path, world, expected):
+    """ws=4 and ws=8 subgroup probes on gloo: the exact subgroup ladder the
+    driver's 8-GPU RCCL run executes (verdict: be ready for the day an
+    8-GPU node appears)."""
+    script = tmp_path / "subgroup_main.py"
+    script.write_text(SUBGROUP_SCRIPT)
+    out = run_torchrun([str(script)], nproc=world, timeout=600, port=29518 + world)
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("SUBGROUP_RESULT")][0]
+    res = json.loads(line.split(" ", 1)[1])
+    assert res["hbm_copy_gbps"] > 0, res
+    assert set(res["busbw"]) == expected, res
+    assert all(v > 0 for v in res["busbw"].values()), res

@@ -159,6 +159,12 @@ def test_rccl_cell_probe_binary():
     res = runner.probe_cell([0], size_mb=16, iters=3)
     assert res["ok"], res
     assert res["ndev"] == 1
+    # single-device probe flags itself as an HBM copy, not an xGMI number
+    assert res["hbm_copy"] is True
+    if torch.cuda.device_count() >= 2:
+        multi = runner.probe_cell([0, 1], size_mb=16, iters=3, p2p_matrix=True)
+        assert multi["ok"] and multi["hbm_copy"] is False, multi
+        assert "0-1" in multi.get("p2p_matrix", {}), multi
 
 
 @needs_gpu
@@ -209,7 +215,10 @@ def test_bench_on_gpu():
     assert out.returncode == 0, out.stderr[-3000:]
     result = json.loads(out.stdout.strip().splitlines()[-1])
     assert result["config"]["vc_safety_violations"] == 0
-    assert result["config"]["rccl_busbw_gbps"], "expected a busbw measurement on GPU"
+    # 1-GPU run: the local-copy probe is reported as hbm_copy_gbps (an HBM
+    # health floor); rccl_busbw_gbps carries only true multi-GPU xGMI numbers
+    assert result["config"]["hbm_copy_gbps"], "expected an HBM copy measurement on GPU"
+    assert result["config"]["rccl_busbw_gbps"] == {}
 
 
 @needs_gpu

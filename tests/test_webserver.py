@@ -229,3 +229,24 @@ def test_event_loop_not_blocked_by_fifo_wait():
     # concurrent requests must not have waited for the 200 ms block
     assert healthz_t < 0.15, f"/healthz stalled {healthz_t * 1e3:.0f} ms behind FIFO block"
     assert health_post_t < 0.15, f"health POST stalled {health_post_t * 1e3:.0f} ms"
+
+
+def test_probe_result_suspect_links_localize(client):
+    """A >2-GPU probe with a p2p matrix marks the localized links, not the
+    probed leaves."""
+    r = client.post("/v1/health/probes", json={
+        "ok": True, "healthy": False, "node": "node1", "group": "g2",
+        "leafCellIndices": [4, 5, 6, 7], "busbw_gbps": 20.0,
+        "suspect_links": [[6, 7, 9.5]]})
+    assert r.status_code == 200
+    links = {(l["a"], l["b"]): l for l in client.get("/v1/inspect/links/node1").json()}
+    assert not links[(6, 7)]["healthy"] and links[(6, 7)]["gbps"] == 9.5
+    status = client.get(constants.PhysicalClusterPath).json()
+
+    def walk(c):
+        yield c
+        for ch in c.get("cellChildren") or []:
+            yield from walk(ch)
+
+    leaves = [c for top in status for c in walk(top) if c.get("leafCellIndex") is not None]
+    assert all(c["cellHealthiness"] == "Healthy" for c in leaves)
