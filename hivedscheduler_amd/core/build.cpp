@@ -99,6 +99,7 @@ struct BuildContext {
                                      cs.address + "'");
       }
       c->totalLeaf = 1;
+      c->freeLeavesUnder = 1;  // all cells start Free
       c->nodes = {ownNodeName};
       c->leafIndices = {leafIndex};
       // measured per-GPU capacity from discovery when provided (a GPU
@@ -119,6 +120,7 @@ struct BuildContext {
         child->parent = c;
         c->children.push_back(child);
         c->totalLeaf += child->totalLeaf;
+        c->freeLeavesUnder += child->freeLeavesUnder;
         c->hbmBytes += child->hbmBytes;
         for (auto& n : child->nodes) {
           if (std::find(c->nodes.begin(), c->nodes.end(), n) == c->nodes.end()) c->nodes.push_back(n);
@@ -148,6 +150,7 @@ struct BuildContext {
     c->preassigned = (preassigned == nullptr) ? c : preassigned;
     if (level == kLowestLevel) {
       c->totalLeaf = 1;
+      c->freeLeavesUnder = 1;  // all cells start Free
     } else {
       int childNum = childCountAt(spec.cellTypes, c->typeName);
       for (int i = 0; i < childNum; i++) {
@@ -156,6 +159,7 @@ struct BuildContext {
         child->parent = c;
         c->children.push_back(child);
         c->totalLeaf += child->totalLeaf;
+        c->freeLeavesUnder += child->freeLeavesUnder;
       }
     }
     fullList.add(c, level);

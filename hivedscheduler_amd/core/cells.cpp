@@ -28,6 +28,11 @@ const char* to_string(GState s) {
 void setCellPriority(Cell* c, int p) {
   int original = c->priority;
   c->priority = p;
+  // maintain the freeLeavesUnder cache on leaf free<->used transitions
+  if (c->level == kLowestLevel && (original == kFreePriority) != (p == kFreePriority)) {
+    int delta = p == kFreePriority ? 1 : -1;
+    for (Cell* a = c; a != nullptr; a = a->parent) a->freeLeavesUnder += delta;
+  }
   if (Cell* parent = c->parent) {
     if (p > parent->priority) {
       setCellPriority(parent, p);

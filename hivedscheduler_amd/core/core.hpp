@@ -78,6 +78,13 @@ struct Cell {
   bool healthy = true;
   // priority -> used leaf-cell count (rolled up the ancestor path)
   std::map<int, int> usedLeafAtPriority;
+  // count of leaves under this cell at kFreePriority, maintained by
+  // setCellPriority on leaf free<->used transitions. Together with
+  // usedLeafAtPriority this gives the placement engine O(#priorities)
+  // availability per node instead of an O(subtree) walk — the filter
+  // hot path's dominant cost at cluster scale (gprof: availLeaves was
+  // 100% of flat samples before this cache).
+  int freeLeavesUnder = 0;
 
   virtual ~Cell() = default;
   virtual bool isPhysical() const = 0;

@@ -56,9 +56,18 @@ void checkInvariants(const HivedCore& core) {
             fail("priority roll-up at " + c->address + ": " + std::to_string(c->priority) +
                  " != max(children) " + std::to_string(maxPrio));
           }
+          int freeSum = 0;
+          for (Cell* childC : c->children) freeSum += childC->freeLeavesUnder;
+          if (c->freeLeavesUnder != freeSum) {
+            fail("freeLeavesUnder roll-up at " + c->address + ": " +
+                 std::to_string(c->freeLeavesUnder) + " != sum(children) " +
+                 std::to_string(freeSum));
+          }
           if (c->healthy != allHealthy) fail("healthiness roll-up at " + c->address);
           if (anyUsed && c->state != CState::Used) fail("state roll-up (Used) at " + c->address);
           if (usedSum != c->usedLeafAtPriority) fail("usedLeafAtPriority roll-up at " + c->address);
+        } else if (c->freeLeavesUnder != (c->priority == kFreePriority ? 1 : 0)) {
+          fail("leaf freeLeavesUnder at " + c->address);
         }
         // binding symmetry
         if (c->virt != nullptr && c->virt->phys != c) fail("asymmetric binding at " + c->address);
@@ -197,8 +206,17 @@ void checkInvariants(const HivedCore& core) {
           }
           if (!v->children.empty()) {
             int maxPrio = kFreePriority;
-            for (Cell* child : v->children) maxPrio = std::max(maxPrio, child->priority);
+            int freeSum = 0;
+            for (Cell* child : v->children) {
+              maxPrio = std::max(maxPrio, child->priority);
+              freeSum += child->freeLeavesUnder;
+            }
             if (v->priority != maxPrio) fail("virtual priority roll-up at " + v->address);
+            if (v->freeLeavesUnder != freeSum) {
+              fail("virtual freeLeavesUnder roll-up at " + v->address);
+            }
+          } else if (v->freeLeavesUnder != (v->priority == kFreePriority ? 1 : 0)) {
+            fail("virtual leaf freeLeavesUnder at " + v->address);
           }
         }
       }

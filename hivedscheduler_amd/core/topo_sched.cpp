@@ -83,12 +83,40 @@ bool leafUsable(Cell* c, const PickSession& s) {
 // (availableTotal, availableFree): usable leaves that are free or
 // lower-priority (preemptible), excluding ones already taken or excluded in
 // this session.
+//
+// O(#priorities) fast path from the freeLeavesUnder / usedLeafAtPriority
+// caches, valid when the session carries no per-leaf state and the subtree
+// has no bad leaves: a PHYSICAL cell's `healthy` roll-up proves that; a
+// bound virtual cell inherits its physical partner's proof; an UNBOUND
+// virtual cell has no bound descendants at all (bindings are upward-
+// contiguous: bindCell binds leaf->root, exposure bindings require a bound
+// parent), so every leaf under it is usable by construction.
 std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
   if (c->level == kLowestLevel) {
-    if (s.taken.count(c) || s.excluded.count(c) || !leafUsable(c, s)) return {0, 0};
+    if ((!s.taken.empty() && s.taken.count(c)) || (!s.excluded.empty() && s.excluded.count(c)) ||
+        !leafUsable(c, s)) {
+      return {0, 0};
+    }
     if (c->priority == kFreePriority) return {1, 1};
     if (c->priority < p) return {1, 0};
     return {0, 0};
+  }
+  if (s.minHbm == 0 && s.taken.empty() && s.excluded.empty()) {
+    bool proven;
+    if (c->isPhysical()) {
+      proven = static_cast<PhysicalCell*>(c)->healthy;
+    } else {
+      PhysicalCell* ph = static_cast<VirtualCell*>(c)->phys;
+      proven = ph == nullptr || ph->healthy;
+    }
+    if (proven) {
+      int preemptible = 0;
+      for (auto& [pp, n] : c->usedLeafAtPriority) {
+        if (pp >= p) break;
+        preemptible += n;
+      }
+      return {c->freeLeavesUnder + preemptible, c->freeLeavesUnder};
+    }
   }
   int at = 0, af = 0;
   for (Cell* child : c->children) {
