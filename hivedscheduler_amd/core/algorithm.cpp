@@ -409,8 +409,32 @@ bool HivedCore::handleSchedulingRequest(const SchedulingRequest& sr, Placement<P
   return scheduleOpportunisticGroup(sr, phys, failedReason);
 }
 
-bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& sr, Placement<PhysicalCell>* phys,
+bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement<PhysicalCell>* phys,
                                         Placement<VirtualCell>* virt, std::string* failedReason) {
+  // When the chain carries degraded xGMI links, feed the virtual descent the
+  // physical clean-shape caps (level -> max link-clean free capacity of any
+  // one physical cell) so its link-honoring attempts pick shapes that admit
+  // a clean mapping; the dirty rungs of the attempt ladder run uncapped.
+  SchedulingRequest sr = srIn;
+  std::map<int, int> cleanCaps;
+  {
+    auto& ccl = fullCellList_[sr.chain];
+    bool chainHasBadLinks = false;
+    for (Cell* c : ccl.at(ccl.top())) {
+      if (static_cast<PhysicalCell*>(c)->badLinksUnder > 0) {
+        chainHasBadLinks = true;
+        break;
+      }
+    }
+    if (chainHasBadLinks) {
+      for (int l = kLowestLevel; l <= ccl.top(); l++) {
+        int best = 0;
+        for (Cell* c : ccl.at(l)) best = std::max(best, cleanFreeLeafCapacity(c));
+        cleanCaps[l] = best;
+      }
+      sr.physCleanCaps = &cleanCaps;
+    }
+  }
   if (!vcSchedulers_[sr.vc].schedule(sr, virt, failedReason)) return false;
 
   std::unordered_map<VirtualCell*, PhysicalCell*> bindings;

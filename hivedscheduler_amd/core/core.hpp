@@ -285,6 +285,15 @@ struct SchedulingRequest {
   const std::set<std::string>* suggestedNodes = nullptr;
   bool ignoreSuggestedNodes = true;
   long long hbmBytes = 0;  // minimum per-leaf HBM capacity (0 = any)
+  // Physical clean-shape feasibility caps, set only when the chain carries
+  // degraded xGMI links: level -> max link-clean FREE leaf capacity of any
+  // single physical cell at that level. The virtual descent (link-blind for
+  // unbound virtual cells) caps per-cell availability with these in its
+  // link-honoring attempts, so a request whose clean mapping needs a
+  // lower-affinity shape (e.g. a quad request that must straddle quads
+  // because of a bad leaf + bad link) picks that shape up front instead of
+  // falling to a dirty placement.
+  const std::map<int, int>* physCleanCaps = nullptr;
 };
 
 // ---------------------------------------------------------------------------
@@ -300,10 +309,12 @@ class TopoScheduler {
 
   // Returns placement or empty with failedReason set. minHbmBytes > 0
   // filters out leaves whose measured HBM capacity falls short.
+  // physCleanCaps (optional): see SchedulingRequest::physCleanCaps.
   bool Schedule(const std::map<int, int>& podLeafCellNums, int priority,
                 const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
                 Placement<Cell>* out, std::string* failedReason,
-                long long minHbmBytes = 0) const;
+                long long minHbmBytes = 0,
+                const std::map<int, int>* physCleanCaps = nullptr) const;
 
  private:
   struct NodeView {
@@ -316,7 +327,8 @@ class TopoScheduler {
   };
   bool tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                              const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
-                             long long minHbmBytes, bool honorLinks, Placement<Cell>* out,
+                             long long minHbmBytes, bool honorLinks,
+                             const std::map<int, int>* physCleanCaps, Placement<Cell>* out,
                              std::string* failedReason) const;
 
   std::vector<Cell*> viewCells_;
@@ -558,6 +570,8 @@ PhysicalCell* findPhysicalLeafCell(std::map<std::string, ChainCellList>& fullCel
                                    const std::string& chain, const std::string& node,
                                    int leafIndex);
 Cell* ancestorNoHigherThanNode(Cell* c);
+// max link-clean FREE leaf capacity within one cell (no session state)
+int cleanFreeLeafCapacity(Cell* c);
 void checkInvariants(const HivedCore& core);
 
 }  // namespace hived

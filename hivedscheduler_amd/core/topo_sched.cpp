@@ -65,6 +65,14 @@ struct PickSession {
   std::unordered_set<Cell*> excluded;
   // minimum per-leaf HBM capacity demanded by the request (0 = any)
   long long minHbm = 0;
+  // link-honoring attempts on VIRTUAL views only: level -> max link-clean
+  // FREE leaf capacity of any single physical cell at that level. An
+  // unbound virtual cell must eventually map onto ONE physical cell of its
+  // level, so its clean availability cannot exceed the best physical
+  // cell's; capping here makes the (otherwise link-blind) virtual descent
+  // choose shapes that admit a clean physical mapping (e.g. straddle quads
+  // when no single quad has 4 clean leaves).
+  const std::map<int, int>* caps = nullptr;
 };
 
 // Leaf-granular availability: a bad physical leaf (sick GPU) is never
@@ -101,7 +109,14 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
     if (c->priority < p) return {1, 0};
     return {0, 0};
   }
-  if (s.minHbm == 0 && s.taken.empty() && s.excluded.empty()) {
+  // physical clean-shape caps apply to unbound virtual cells; capping is
+  // RECURSIVE (parent sums capped children, then caps itself), so whenever
+  // caps are active the counter fast path is skipped EVERYWHERE — a bound
+  // parent's uncapped counters would overstate what its capped unbound
+  // children can deliver and underflow the drain in pickLeaves
+  bool capped = s.caps != nullptr && !c->isPhysical() &&
+                static_cast<VirtualCell*>(c)->phys == nullptr;
+  if (s.caps == nullptr && s.minHbm == 0 && s.taken.empty() && s.excluded.empty()) {
     bool proven;
     if (c->isPhysical()) {
       proven = static_cast<PhysicalCell*>(c)->healthy;
@@ -123,6 +138,13 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
     auto [t, f] = availLeaves(child, p, s);
     at += t;
     af += f;
+  }
+  if (capped) {
+    auto it = s.caps->find(c->level);
+    if (it != s.caps->end()) {
+      at = std::min(at, it->second);
+      af = std::min(af, it->second);
+    }
   }
   return {at, af};
 }
@@ -323,10 +345,19 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 
 }  // namespace
 
+// Max link-clean FREE leaf capacity within one cell: free healthy leaves
+// minus the minimum exclusions needed so no degraded link has both
+// endpoints in the set. Used to build SchedulingRequest::physCleanCaps.
+int cleanFreeLeafCapacity(Cell* c) {
+  PickSession empty;
+  return cleanAvailAnalysis(c, kOpportunisticPriority, empty, INT_MAX, nullptr);
+}
+
 bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                                           const std::set<std::string>& suggestedNodes,
                                           bool ignoreSuggestedNodes, long long minHbmBytes,
-                                          bool honorLinks, Placement<Cell>* out,
+                                          bool honorLinks, const std::map<int, int>* physCleanCaps,
+                                          Placement<Cell>* out,
                                           std::string* failedReason) const {
   // Build and sort the cluster view: healthy > suggested > same-priority used
   // (desc, packing) > higher-priority used (asc, stay away).
@@ -334,6 +365,7 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   cv.reserve(viewCells_.size());
   PickSession probe;  // empty: availability before any placement
   probe.minHbm = minHbmBytes;
+  if (honorLinks) probe.caps = physCleanCaps;
   for (Cell* c : viewCells_) {
     NodeView n;
     n.c = c;
@@ -421,6 +453,7 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   PickSession session;
   session.minHbm = minHbmBytes;
   if (honorLinks) {
+    session.caps = physCleanCaps;
     // Gang-wide link cleanliness per node: every pod of the gang on one node
     // communicates with every other (the gang's collective), so exclusions
     // are computed from the node's TOTAL gang demand before any pick.
@@ -447,14 +480,18 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
 bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int priority,
                              const std::set<std::string>& suggestedNodes,
                              bool ignoreSuggestedNodes, Placement<Cell>* out,
-                             std::string* failedReason, long long minHbmBytes) const {
+                             std::string* failedReason, long long minHbmBytes,
+                             const std::map<int, int>* physCleanCaps) const {
   std::vector<int> sortedLeafNums;
   for (auto& [leafNum, podNum] : podLeafCellNums) {
     for (int i = 0; i < podNum; i++) sortedLeafNums.push_back(leafNum);
   }
-  bool anyBadLinks = false;
+  // caps are only passed when the chain carries degraded links, so their
+  // presence alone demands the dirty-retry rungs of the ladder (virtual
+  // views are mostly unbound and show badLinksUnderOf == 0)
+  bool anyBadLinks = physCleanCaps != nullptr;
   for (Cell* c : viewCells_) {
-    if (badLinksUnderOf(c) > 0) {
+    if (anyBadLinks || badLinksUnderOf(c) > 0) {
       anyBadLinks = true;
       break;
     }
@@ -465,7 +502,7 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
   // degraded link (slow xGMI), hence free+dirty before preempt+clean.
   auto attempt = [&](int p, bool honorLinks) {
     return tryScheduleAtPriority(sortedLeafNums, p, suggestedNodes, ignoreSuggestedNodes,
-                                 minHbmBytes, honorLinks, out, failedReason);
+                                 minHbmBytes, honorLinks, physCleanCaps, out, failedReason);
   };
   if (attempt(kOpportunisticPriority, true)) return true;
   if (anyBadLinks && attempt(kOpportunisticPriority, false)) return true;
@@ -489,7 +526,8 @@ bool IntraVCScheduler::schedule(const SchedulingRequest& sr, Placement<VirtualCe
   Placement<Cell> generic;
   if (scheduler == nullptr ||
       !scheduler->Schedule(sr.podLeafCellNums, sr.priority, *sr.suggestedNodes,
-                           sr.ignoreSuggestedNodes, &generic, failedReason, sr.hbmBytes)) {
+                           sr.ignoreSuggestedNodes, &generic, failedReason, sr.hbmBytes,
+                           sr.physCleanCaps)) {
     if (failedReason->empty()) *failedReason = "no scheduler for request";
     *failedReason += " when scheduling in VC " + sr.vc;
     return false;

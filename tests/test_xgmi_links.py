@@ -196,15 +196,13 @@ def test_config_supplied_link_table():
 
 
 def test_bad_leaf_and_bad_link_compose():
-    """Leaf badness and link degradation are independent dimensions.
-
-    KNOWN BOUND (documented in docs/design.md): a guaranteed 4-GPU request
-    here can only be link-clean by splitting across quads ({2,3}+{4,5}), but
-    the virtual descent (link-blind for unbound virtual cells) commits to a
-    whole quad and the topology-preserving mapping cannot split it, so the
-    dirty fallback places [0,1,2,3]. The hard guarantees that DO hold: the
-    bad GPU is never used, and the request binds. An OPPORTUNISTIC request
-    (physical view, fully link-aware) places clean."""
+    """Leaf badness and link degradation are independent dimensions, and
+    they COMPOSE on the guaranteed path: with GPU 7 bad and link 0<->1
+    degraded, no single quad has 4 clean GPUs, so a clean placement must
+    straddle quads. The physical clean-shape caps fed to the virtual
+    descent (SchedulingRequest::physCleanCaps) make it pick that shape:
+    the topology-preserving mapping then lands e.g. [0,4,5,6] — clean AND
+    avoiding the bad GPU."""
     sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
     sim.alg.set_leaf_cell_healthy("node1", 7, False)
     _mark_link(sim, "node1", 0, 1, False)
@@ -212,15 +210,28 @@ def test_bad_leaf_and_bad_link_compose():
     assert r.kind == "bind"
     cells = set(r.bind_info.leafCellIsolation)
     assert 7 not in cells
+    assert not ({0, 1} <= cells), sorted(cells)
     sim.alg._core.check_invariants()
     sim.delete_pod("ns/p1")
-    # the opportunistic path schedules on the physical view where link state
-    # is fully visible: it places clean (one endpoint dropped)
+    # the opportunistic path (physical view, fully link-aware) places clean
     r2 = sim.schedule("ns/p2", sim.pod_spec(leaf_cells=4, priority=-1))
     assert r2.kind == "bind"
     cells2 = set(r2.bind_info.leafCellIsolation)
     assert 7 not in cells2
     assert not ({0, 1} <= cells2), sorted(cells2)
+    sim.alg._core.check_invariants()
+
+
+def test_clean_shape_caps_dont_break_capacity():
+    """Caps only shape link-honoring attempts: with link 0<->1 degraded,
+    two 4-GPU guaranteed jobs still fill the whole node (first clean quad,
+    then the dirty remainder)."""
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    _mark_link(sim, "node1", 0, 1, False)
+    r1 = sim.schedule("ns/p1", sim.pod_spec(leaf_cells=4))
+    assert sorted(r1.bind_info.leafCellIsolation) == [4, 5, 6, 7]
+    r2 = sim.schedule("ns/p2", sim.pod_spec(leaf_cells=4))
+    assert sorted(r2.bind_info.leafCellIsolation) == [0, 1, 2, 3]
     sim.alg._core.check_invariants()
 
 
