@@ -92,7 +92,12 @@ def config2(iters, gpu):
                 if size <= n:
                     p = runner.probe_cell(list(range(size)), size_mb=64, iters=10)
                     if p.get("ok"):
-                        curve[str(size)] = round(p["busbw_gbps"], 2)
+                        if size == 1:
+                            # a 1-device probe never leaves the GPU: it is an
+                            # HBM copy floor, not an xGMI number
+                            out["hbm_copy_gbps"] = round(p["busbw_gbps"], 2)
+                        else:
+                            curve[str(size)] = round(p["busbw_gbps"], 2)
             out["rccl_busbw_gbps"] = curve
     return out
 
