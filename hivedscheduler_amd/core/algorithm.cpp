@@ -427,15 +427,34 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
       }
     }
     if (chainHasBadLinks) {
+      // honor the suggested-node restriction the mapping stage will apply:
+      // capacity on non-suggested nodes cannot satisfy this request
+      const std::set<std::string>* filter =
+          sr.ignoreSuggestedNodes ? nullptr : sr.suggestedNodes;
       for (int l = kLowestLevel; l <= ccl.top(); l++) {
         int best = 0;
-        for (Cell* c : ccl.at(l)) best = std::max(best, cleanFreeLeafCapacity(c));
+        for (Cell* c : ccl.at(l)) best = std::max(best, cleanFreeLeafCapacity(c, filter));
         cleanCaps[l] = best;
       }
       sr.physCleanCaps = &cleanCaps;
+      if (mapDebugRelease()) {
+        fprintf(stderr, "[caps] chain=%s ignoreSug=%d:", sr.chain.c_str(),
+                (int)sr.ignoreSuggestedNodes);
+        for (auto& [l, v] : cleanCaps) fprintf(stderr, " L%d=%d", l, v);
+        fprintf(stderr, "\n");
+      }
     }
   }
   if (!vcSchedulers_[sr.vc].schedule(sr, virt, failedReason)) return false;
+  if (mapDebugRelease()) {
+    for (auto& [ln, pods] : *virt) {
+      for (auto& pod : pods) {
+        fprintf(stderr, "[virt ln=%d]", ln);
+        for (auto* v : pod) fprintf(stderr, " %s", v->address.c_str());
+        fprintf(stderr, "\n");
+      }
+    }
+  }
 
   std::unordered_map<VirtualCell*, PhysicalCell*> bindings;
   auto lazyPreempted = tryLazyPreempt(*virt, sr.groupName);

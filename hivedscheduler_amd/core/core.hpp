@@ -570,8 +570,9 @@ PhysicalCell* findPhysicalLeafCell(std::map<std::string, ChainCellList>& fullCel
                                    const std::string& chain, const std::string& node,
                                    int leafIndex);
 Cell* ancestorNoHigherThanNode(Cell* c);
-// max link-clean FREE leaf capacity within one cell (no session state)
-int cleanFreeLeafCapacity(Cell* c);
+// max link-clean FREE leaf capacity within one cell (no session state);
+// suggestedNodes non-null restricts usable leaves to those nodes
+int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes = nullptr);
 void checkInvariants(const HivedCore& core);
 
 }  // namespace hived

@@ -65,6 +65,9 @@ struct PickSession {
   std::unordered_set<Cell*> excluded;
   // minimum per-leaf HBM capacity demanded by the request (0 = any)
   long long minHbm = 0;
+  // restrict leaf usability to these nodes (caps computation honors the
+  // K8s suggested-node set the mapping stage will enforce)
+  const std::set<std::string>* suggestedFilter = nullptr;
   // link-honoring attempts on VIRTUAL views only: level -> max link-clean
   // FREE leaf capacity of any single physical cell at that level. An
   // unbound virtual cell must eventually map onto ONE physical cell of its
@@ -85,6 +88,10 @@ bool leafUsable(Cell* c, const PickSession& s) {
   if (p == nullptr) return true;  // unbound virtual: no physical facts yet
   if (!p->healthy) return false;
   if (s.minHbm > 0 && p->hbmBytes > 0 && p->hbmBytes < s.minHbm) return false;
+  if (s.suggestedFilter != nullptr &&
+      (p->nodes.empty() || !s.suggestedFilter->count(p->nodes[0]))) {
+    return false;
+  }
   return true;
 }
 
@@ -116,7 +123,8 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
   // children can deliver and underflow the drain in pickLeaves
   bool capped = s.caps != nullptr && !c->isPhysical() &&
                 static_cast<VirtualCell*>(c)->phys == nullptr;
-  if (s.caps == nullptr && s.minHbm == 0 && s.taken.empty() && s.excluded.empty()) {
+  if (s.caps == nullptr && s.minHbm == 0 && s.suggestedFilter == nullptr && s.taken.empty() &&
+      s.excluded.empty()) {
     bool proven;
     if (c->isPhysical()) {
       proven = static_cast<PhysicalCell*>(c)->healthy;
@@ -348,8 +356,9 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 // Max link-clean FREE leaf capacity within one cell: free healthy leaves
 // minus the minimum exclusions needed so no degraded link has both
 // endpoints in the set. Used to build SchedulingRequest::physCleanCaps.
-int cleanFreeLeafCapacity(Cell* c) {
+int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes) {
   PickSession empty;
+  empty.suggestedFilter = suggestedNodes;
   return cleanAvailAnalysis(c, kOpportunisticPriority, empty, INT_MAX, nullptr);
 }
 

@@ -14,8 +14,13 @@ Scenario YAML:
     - delete: {pod: a/train-0}
     - nodeHealth: {node: node2, healthy: false}
     - leafHealth: {node: node1, leafIndex: 3, healthy: true}
+    - linkHealth: {node: node1, a: 0, b: 1, healthy: false, gbps: 12.5}
     - preemptToCompletion: {pod: b/prod-0, vc: prod, priority: 200,
                             leafCellNumber: 8}
+
+Submit events may also assert `expectCellsNot: [0, 1]` (the placement must
+not contain ALL of the listed GPU indices — e.g. both endpoints of a
+degraded link) and request `hbmBytesPerCell`.
 
 Exit code 0 when every `expect` matched, 1 otherwise. This is the same
 in-memory path the test suite and bench.py drive (SURVEY.md §4's K8s-free
@@ -42,6 +47,7 @@ def build_spec(ev: dict) -> PodSchedulingSpec:
         leafCellNumber=int(ev.get("leafCellNumber", 1)),
         lazyPreemptionEnable=bool(ev.get("lazyPreemptionEnable", False)),
         ignoreK8sSuggestedNodes=bool(ev.get("ignoreK8sSuggestedNodes", True)),
+        hbmBytesPerCell=int(ev.get("hbmBytesPerCell", 0)),
     )
     if ev.get("group"):
         spec.affinityGroup = AffinityGroupSpec(
@@ -99,6 +105,12 @@ def main(argv=None) -> int:
                 if want_node and (r.kind != "bind" or r.bind_info.node != want_node):
                     print(f"    EXPECT FAILED: wanted node {want_node}", file=sys.stderr)
                     failures += 1
+                cells_not = event.get("expectCellsNot")
+                if cells_not and r.kind == "bind" and set(
+                        int(c) for c in cells_not) <= set(r.bind_info.leafCellIsolation):
+                    print(f"    EXPECT FAILED: placement contains all of {cells_not} "
+                          "(e.g. both endpoints of a degraded link)", file=sys.stderr)
+                    failures += 1
             elif "delete" in event:
                 sim.delete_pod(event["delete"]["pod"])
                 print(f"[{i}] deleted {event['delete']['pod']}")
@@ -111,6 +123,13 @@ def main(argv=None) -> int:
                 sim.alg.set_leaf_cell_healthy(ev["node"], int(ev["leafIndex"]),
                                               bool(ev["healthy"]))
                 print(f"[{i}] leaf {ev['node']}/{ev['leafIndex']} healthy={ev['healthy']}")
+            elif "linkHealth" in event:
+                ev = event["linkHealth"]
+                sim.alg.set_xgmi_link_healthy(ev["node"], int(ev["a"]), int(ev["b"]),
+                                              bool(ev["healthy"]),
+                                              float(ev.get("gbps", 0.0)))
+                print(f"[{i}] link {ev['node']}/{ev['a']}<->{ev['b']} "
+                      f"healthy={ev['healthy']}")
             else:
                 print(f"[{i}] unknown event {sorted(event)}", file=sys.stderr)
                 failures += 1

@@ -87,3 +87,18 @@ def test_scenario_cli():
         capture_output=True, text=True, timeout=120, cwd=REPO)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "0 expectation failure(s)" in out.stdout
+
+
+def test_scenario_degraded_link_cli():
+    """The degraded-link scenario: gangs avoid the bad 0<->1 link under a
+    suggested-node restriction, 1-GPU work still uses the endpoints, the
+    healed link serves gangs again (exit 0)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "hivedscheduler_amd.sim",
+         os.path.join(REPO, "examples/scenario-degraded-link.yaml")],
+        capture_output=True, text=True, timeout=120, cwd=REPO)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "0 expectation failure(s)" in out.stdout

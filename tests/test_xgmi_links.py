@@ -279,3 +279,29 @@ def test_hbm_negative_demand_rejected():
     sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
     with pytest.raises(WebServerError):
         sim.schedule("ns/p1", sim.pod_spec(leaf_cells=1, hbm_bytes_per_cell=-1))
+
+
+def test_clean_caps_honor_suggested_nodes():
+    """Regression: the clean-shape caps must be computed over SUGGESTED
+    nodes only — capacity on non-suggested nodes cannot satisfy the request,
+    and counting it made the virtual descent pick unmappable-clean shapes
+    that fell through to dirty placements (found via the degraded-link
+    example scenario on the rack-chain config)."""
+    sim = SimScheduler(mi355x_cluster_config(
+        num_nodes=2, vcs={"VC1": [("MI355X-NODE", 2)]}))
+    _mark_link(sim, "node1", 0, 1, False)
+    sug = ["node1"]
+    r1 = sim.schedule("ns/pair", sim.pod_spec(leaf_cells=2, ignore_suggested=False),
+                      suggested=sug)
+    assert r1.bind_info.node == "node1"
+    assert sorted(r1.bind_info.leafCellIsolation) != [0, 1]
+    # 4 GPUs with pairs [4,5] (wait, the pair landed on a clean pair) used:
+    # no single quad of node1 has 4 clean free GPUs once its clean quad is
+    # partially used -> the clean shape must straddle quads ON NODE1 (node2
+    # capacity must not fool the caps)
+    r2 = sim.schedule("ns/quad", sim.pod_spec(leaf_cells=4, ignore_suggested=False),
+                      suggested=sug)
+    assert r2.kind == "bind" and r2.bind_info.node == "node1"
+    cells = set(r2.bind_info.leafCellIsolation)
+    assert not ({0, 1} <= cells), sorted(cells)
+    sim.alg._core.check_invariants()
