@@ -416,7 +416,7 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
   // one physical cell) so its link-honoring attempts pick shapes that admit
   // a clean mapping; the dirty rungs of the attempt ladder run uncapped.
   SchedulingRequest sr = srIn;
-  std::map<int, int> cleanCaps;
+  CleanShapeWorld cleanWorld;
   {
     auto& ccl = fullCellList_[sr.chain];
     bool chainHasBadLinks = false;
@@ -426,21 +426,21 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
         break;
       }
     }
-    if (chainHasBadLinks) {
+    int gangLeaves = 0;
+    for (auto& [ln, pn] : sr.podLeafCellNums) gangLeaves += ln * pn;
+    // a 1-leaf request never straddles a link: no world (it may freely use
+    // — and by the parking preference, should use — degraded endpoints)
+    if (chainHasBadLinks && gangLeaves >= 2) {
       // honor the suggested-node restriction the mapping stage will apply:
       // capacity on non-suggested nodes cannot satisfy this request
       const std::set<std::string>* filter =
           sr.ignoreSuggestedNodes ? nullptr : sr.suggestedNodes;
-      for (int l = kLowestLevel; l <= ccl.top(); l++) {
-        int best = 0;
-        for (Cell* c : ccl.at(l)) best = std::max(best, cleanFreeLeafCapacity(c, filter));
-        cleanCaps[l] = best;
-      }
-      sr.physCleanCaps = &cleanCaps;
+      cleanWorld = computeCleanShapeWorld(ccl, filter);
+      sr.cleanWorld = &cleanWorld;
       if (mapDebugRelease()) {
-        fprintf(stderr, "[caps] chain=%s ignoreSug=%d:", sr.chain.c_str(),
-                (int)sr.ignoreSuggestedNodes);
-        for (auto& [l, v] : cleanCaps) fprintf(stderr, " L%d=%d", l, v);
+        fprintf(stderr, "[world] chain=%s ignoreSug=%d excl=%zu:", sr.chain.c_str(),
+                (int)sr.ignoreSuggestedNodes, cleanWorld.excluded.size());
+        for (auto& [l, v] : cleanWorld.caps) fprintf(stderr, " L%d=%d", l, v);
         fprintf(stderr, "\n");
       }
     }
@@ -535,7 +535,7 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
     return mapVirtualPlacementToPhysical(preassigned, nonPreassigned,
                                          freeCellList_[sr.chain].shallowCopy(), freeCellNumCopy,
                                          *sr.suggestedNodes, sr.ignoreSuggestedNodes, bindings,
-                                         sr.hbmBytes, honorLinks);
+                                         sr.hbmBytes, honorLinks, sr.cleanWorld);
   };
   bool mapped = anyBadLinks ? (tryMap(true) || tryMap(false)) : tryMap(false);
   if (mapped) {

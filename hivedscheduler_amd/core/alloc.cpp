@@ -17,16 +17,46 @@ static bool mapDebug() {
 
 namespace {
 
+// Refcounted claims with an undo journal: physical leaves (and their
+// ancestors) claimed by picks of THIS mapping round. The journal makes
+// backtracking EXACT — abandoned branches are fully unclaimed. That matters
+// twice: stale claims used to merely bias the packing sort, but with the
+// honorLinks hard constraint a stale claim on a degraded link's endpoint
+// wrongly vetoes its peer on a later branch and turns findable clean
+// mappings into dirty fallbacks (found by the dirty-only-when-forced
+// property oracle, seed 1999-class: the 4-leaf vertex retried on another
+// quad still saw its abandoned quad's claims).
+struct ClaimSet {
+  std::unordered_map<PhysicalCell*, int> refs;
+  std::vector<PhysicalCell*> log;  // claimed leaves, in order
+
+  bool count(PhysicalCell* c) const { return refs.count(c) > 0; }
+  size_t checkpoint() const { return log.size(); }
+  void claim(PhysicalCell* leaf) {
+    log.push_back(leaf);
+    for (PhysicalCell* a = leaf; a != nullptr; a = static_cast<PhysicalCell*>(a->parent)) {
+      refs[a]++;
+    }
+  }
+  void unwindTo(size_t cp) {
+    while (log.size() > cp) {
+      PhysicalCell* leaf = log.back();
+      log.pop_back();
+      for (PhysicalCell* a = leaf; a != nullptr; a = static_cast<PhysicalCell*>(a->parent)) {
+        auto it = refs.find(a);
+        if (--it->second == 0) refs.erase(it);
+      }
+    }
+  }
+};
+
 struct AllocCtx {
   const std::set<std::string>& suggested;
   bool ignoreSuggested;
   std::unordered_map<VirtualCell*, PhysicalCell*>& bindings;
-  // physical cells (and their ancestors) claimed by earlier picks of THIS
-  // mapping round: lets the buddy-packing sort co-locate a group's cells
-  // (e.g. both pairs of a 2-pair group into ONE quad) so intact higher-level
-  // free cells — other VCs' guarantees — survive. Backtracking may leave
-  // stale entries; that only biases packing, never correctness.
-  std::unordered_set<PhysicalCell*>& claimed;
+  // exact claim tracking (see ClaimSet): drives the packing sort keys AND
+  // the honorLinks pick-time link constraint
+  ClaimSet& claimed;
   // minimum measured HBM per leaf (0 = any): leaves below are unusable
   long long minHbm = 0;
   // when true, a leaf candidate whose degraded-link peer is already claimed
@@ -34,6 +64,12 @@ struct AllocCtx {
   // sick link); the backtracking search then finds a clean mapping or fails,
   // and the caller retries with honorLinks=false (capacity over quality)
   bool honorLinks = false;
+  // the clean-shape world the virtual placement was computed in (honorLinks
+  // runs): its excluded leaves are unusable here too, keeping the mapper's
+  // choices consistent with the shape — without this, a mapper pick of an
+  // excluded endpoint forces later cross-group conflicts the per-group
+  // backtracking cannot undo
+  const CleanShapeWorld* world = nullptr;
 };
 
 // Number of gang leaves that will land under this vertex's physical binding
@@ -45,11 +81,7 @@ int vertexLeafDemand(const BindingVertex* v) {
   return n;
 }
 
-void markClaimed(const AllocCtx& ctx, PhysicalCell* c) {
-  for (PhysicalCell* a = c; a != nullptr; a = static_cast<PhysicalCell*>(a->parent)) {
-    ctx.claimed.insert(a);
-  }
-}
+void markClaimed(const AllocCtx& ctx, PhysicalCell* c) { ctx.claimed.claim(c); }
 
 // honorLinks hard constraint, checked at PICK time (claims accumulate while
 // the backtracking search runs): a leaf whose degraded-link peer is already
@@ -76,6 +108,10 @@ std::vector<PhysicalCell*> getUsablePhysicalCells(const std::vector<Cell*>& cand
     if (ctx.minHbm > 0 && c->level == kLowestLevel && c->hbmBytes > 0 &&
         c->hbmBytes < ctx.minHbm) {
       continue;  // leaf's measured HBM falls short of the request's demand
+    }
+    if (ctx.honorLinks && ctx.world != nullptr && c->level == kLowestLevel &&
+        ctx.world->excluded.count(c)) {
+      continue;  // endpoint the clean-shape world avoids
     }
     if (!ctx.ignoreSuggested) {
       bool anySuggested = false;
@@ -171,6 +207,7 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cellsIn,
   int n = static_cast<int>(cells.size());
   int m = static_cast<int>(candidates.size());
   std::vector<int> pickedIdx(n, 0);
+  std::vector<size_t> claimCp(n, 0);  // claims checkpoint before each vertex's pick
   std::vector<bool> used(m, false);
   int cellIndex = 0;
   while (cellIndex >= 0) {
@@ -178,6 +215,7 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cellsIn,
     for (candidateIndex = pickedIdx[cellIndex]; candidateIndex < m; candidateIndex++) {
       if (used[candidateIndex]) continue;
       PhysicalCell* candidate = candidates[candidateIndex];
+      size_t attemptCp = ctx.claimed.checkpoint();
       bool ok;
       if (candidate->level == kLowestLevel) {
         if (leafLinkConflict(ctx, candidate)) continue;
@@ -189,9 +227,13 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cellsIn,
         childVerts.reserve(cells[cellIndex]->children.size());
         for (auto& ch : cells[cellIndex]->children) childVerts.push_back(ch.get());
         ok = mapVirtualCellsToPhysical(childVerts, candidate->children, ctx, false, nullptr);
+        // abandoned branch: unwind its claims exactly, or the honorLinks
+        // link constraint would see ghosts on the next branch
+        if (!ok) ctx.claimed.unwindTo(attemptCp);
       }
       if (ok) {
         pickedIdx[cellIndex] = candidateIndex;
+        claimCp[cellIndex] = attemptCp;
         used[candidateIndex] = true;
         if (cellIndex == n - 1) {
           if (returnPicked) {
@@ -207,6 +249,7 @@ bool mapVirtualCellsToPhysical(const std::vector<BindingVertex*>& cellsIn,
       cellIndex--;
       if (cellIndex >= 0) {
         used[pickedIdx[cellIndex]] = false;
+        ctx.claimed.unwindTo(claimCp[cellIndex]);
         pickedIdx[cellIndex]++;
       }
     } else {
@@ -327,9 +370,10 @@ bool HivedCore::mapVirtualPlacementToPhysical(
     ChainCellList freeList, std::map<int, int> freeCellNum,
     const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
     std::unordered_map<VirtualCell*, PhysicalCell*>& bindings, long long minHbmBytes,
-    bool honorLinks) {
-  std::unordered_set<PhysicalCell*> claimed;
-  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings, claimed, minHbmBytes, honorLinks};
+    bool honorLinks, const CleanShapeWorld* world) {
+  ClaimSet claimed;
+  AllocCtx ctx{suggestedNodes, ignoreSuggestedNodes, bindings, claimed,
+               minHbmBytes, honorLinks, honorLinks ? world : nullptr};
   // pre-claim physical leaves ALREADY bound for this placement (the caller
   // seeds `bindings` with bound virtual leaves, which skip the vertex trees)
   // so the link constraint sees the whole gang, not just the new picks

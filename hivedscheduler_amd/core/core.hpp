@@ -275,6 +275,15 @@ struct ScheduleResult {
 template <class CellT>
 using Placement = std::map<int, std::vector<std::vector<CellT*>>>;
 
+// One globally consistent link-clean view of a chain's free capacity:
+// `excluded` = the physical leaves dropped (per-node max independent set of
+// the degraded-link graph restricted to free leaves); `caps` = per-level max
+// in-world free capacity of any single physical cell.
+struct CleanShapeWorld {
+  std::map<int, int> caps;
+  std::unordered_set<PhysicalCell*> excluded;
+};
+
 struct SchedulingRequest {
   std::string vc;
   std::string pinnedCellId;
@@ -285,15 +294,16 @@ struct SchedulingRequest {
   const std::set<std::string>* suggestedNodes = nullptr;
   bool ignoreSuggestedNodes = true;
   long long hbmBytes = 0;  // minimum per-leaf HBM capacity (0 = any)
-  // Physical clean-shape feasibility caps, set only when the chain carries
-  // degraded xGMI links: level -> max link-clean FREE leaf capacity of any
-  // single physical cell at that level. The virtual descent (link-blind for
-  // unbound virtual cells) caps per-cell availability with these in its
-  // link-honoring attempts, so a request whose clean mapping needs a
-  // lower-affinity shape (e.g. a quad request that must straddle quads
-  // because of a bad leaf + bad link) picks that shape up front instead of
-  // falling to a dirty placement.
-  const std::map<int, int>* physCleanCaps = nullptr;
+  // Clean-shape world, set only when the chain carries degraded xGMI links
+  // and the gang needs >= 2 leaves: ONE globally consistent choice of
+  // link endpoints to avoid. The virtual descent's link-honoring attempts
+  // schedule inside this world (excluded leaves unavailable; unbound
+  // subtrees capped by the best physical cell's in-world capacity; bound
+  // subtrees min-capped by their physical region's in-world capacity), so
+  // a request whose clean mapping needs a lower-affinity shape picks that
+  // shape up front instead of falling to a dirty placement. Dirty rungs
+  // run without the world, so capacity is never sacrificed.
+  const CleanShapeWorld* cleanWorld = nullptr;
 };
 
 // ---------------------------------------------------------------------------
@@ -309,12 +319,12 @@ class TopoScheduler {
 
   // Returns placement or empty with failedReason set. minHbmBytes > 0
   // filters out leaves whose measured HBM capacity falls short.
-  // physCleanCaps (optional): see SchedulingRequest::physCleanCaps.
+  // cleanWorld (optional): see SchedulingRequest::cleanWorld.
   bool Schedule(const std::map<int, int>& podLeafCellNums, int priority,
                 const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
                 Placement<Cell>* out, std::string* failedReason,
                 long long minHbmBytes = 0,
-                const std::map<int, int>* physCleanCaps = nullptr) const;
+                const CleanShapeWorld* cleanWorld = nullptr) const;
 
  private:
   struct NodeView {
@@ -328,7 +338,7 @@ class TopoScheduler {
   bool tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                              const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
                              long long minHbmBytes, bool honorLinks,
-                             const std::map<int, int>* physCleanCaps, Placement<Cell>* out,
+                             const CleanShapeWorld* cleanWorld, Placement<Cell>* out,
                              std::string* failedReason) const;
 
   std::vector<Cell*> viewCells_;
@@ -513,7 +523,8 @@ class HivedCore {
                                      const std::set<std::string>& suggestedNodes,
                                      bool ignoreSuggestedNodes,
                                      std::unordered_map<VirtualCell*, PhysicalCell*>& bindings,
-                                     long long minHbmBytes = 0, bool honorLinks = false);
+                                     long long minHbmBytes = 0, bool honorLinks = false,
+                                     const CleanShapeWorld* world = nullptr);
 
  public:
   // state (public for inspect/bindings simplicity; external mutation forbidden)
@@ -573,6 +584,9 @@ Cell* ancestorNoHigherThanNode(Cell* c);
 // max link-clean FREE leaf capacity within one cell (no session state);
 // suggestedNodes non-null restricts usable leaves to those nodes
 int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes = nullptr);
+// the chain's clean-shape world under one consistent endpoint choice
+CleanShapeWorld computeCleanShapeWorld(const ChainCellList& ccl,
+                                       const std::set<std::string>* suggestedNodes);
 void checkInvariants(const HivedCore& core);
 
 }  // namespace hived

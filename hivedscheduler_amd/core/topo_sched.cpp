@@ -68,15 +68,38 @@ struct PickSession {
   // restrict leaf usability to these nodes (caps computation honors the
   // K8s suggested-node set the mapping stage will enforce)
   const std::set<std::string>* suggestedFilter = nullptr;
-  // link-honoring attempts on VIRTUAL views only: level -> max link-clean
-  // FREE leaf capacity of any single physical cell at that level. An
-  // unbound virtual cell must eventually map onto ONE physical cell of its
-  // level, so its clean availability cannot exceed the best physical
-  // cell's; capping here makes the (otherwise link-blind) virtual descent
-  // choose shapes that admit a clean physical mapping (e.g. straddle quads
-  // when no single quad has 4 clean leaves).
-  const std::map<int, int>* caps = nullptr;
+  // link-honoring attempts on VIRTUAL views only: the chain's clean-shape
+  // world. Excluded physical leaves are unavailable (applied to bound
+  // leaves directly and to bound regions via a physical min-cap); unbound
+  // virtual cells — which must eventually map onto ONE physical cell of
+  // their level — are capped by the best physical cell's in-world
+  // capacity. This makes the (otherwise link-blind) virtual descent choose
+  // shapes that admit a clean physical mapping (e.g. straddle quads when
+  // no single quad has 4 clean leaves).
+  const CleanShapeWorld* world = nullptr;
 };
+
+// in-world free capacity of a PHYSICAL region (free + healthy + not excluded)
+int physFreeInWorld(PhysicalCell* pc, const CleanShapeWorld& w) {
+  if (pc->level == kLowestLevel) {
+    return (pc->priority == kFreePriority && pc->healthy && !w.excluded.count(pc)) ? 1 : 0;
+  }
+  int n = 0;
+  for (Cell* child : pc->children) n += physFreeInWorld(static_cast<PhysicalCell*>(child), w);
+  return n;
+}
+
+// best in-world capacity of any single UNBOUND physical cell at `level`
+// under region `pc` (an unbound virtual cell whose nearest bound ancestor
+// is bound to pc will map onto one of these)
+int maxInWorldAtLevelUnder(PhysicalCell* pc, int level, const CleanShapeWorld& w) {
+  if (pc->level == level) return pc->virt == nullptr ? physFreeInWorld(pc, w) : 0;
+  int best = 0;
+  for (Cell* child : pc->children) {
+    best = std::max(best, maxInWorldAtLevelUnder(static_cast<PhysicalCell*>(child), level, w));
+  }
+  return best;
+}
 
 // Leaf-granular availability: a bad physical leaf (sick GPU) is never
 // available; a virtual leaf bound to a bad physical leaf (doomed-bad binding)
@@ -92,6 +115,7 @@ bool leafUsable(Cell* c, const PickSession& s) {
       (p->nodes.empty() || !s.suggestedFilter->count(p->nodes[0]))) {
     return false;
   }
+  if (s.world != nullptr && s.world->excluded.count(p)) return false;
   return true;
 }
 
@@ -116,14 +140,13 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
     if (c->priority < p) return {1, 0};
     return {0, 0};
   }
-  // physical clean-shape caps apply to unbound virtual cells; capping is
-  // RECURSIVE (parent sums capped children, then caps itself), so whenever
-  // caps are active the counter fast path is skipped EVERYWHERE — a bound
-  // parent's uncapped counters would overstate what its capped unbound
-  // children can deliver and underflow the drain in pickLeaves
-  bool capped = s.caps != nullptr && !c->isPhysical() &&
-                static_cast<VirtualCell*>(c)->phys == nullptr;
-  if (s.caps == nullptr && s.minHbm == 0 && s.suggestedFilter == nullptr && s.taken.empty() &&
+  // the clean-shape world constrains virtual subtrees RECURSIVELY (parent
+  // sums constrained children, then caps itself), so whenever it is active
+  // the counter fast path is skipped EVERYWHERE — a parent's unconstrained
+  // counters would overstate what its constrained children can deliver and
+  // underflow the drain in pickLeaves
+  bool inWorld = s.world != nullptr && !c->isPhysical();
+  if (s.world == nullptr && s.minHbm == 0 && s.suggestedFilter == nullptr && s.taken.empty() &&
       s.excluded.empty()) {
     bool proven;
     if (c->isPhysical()) {
@@ -147,11 +170,41 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
     at += t;
     af += f;
   }
-  if (capped) {
-    auto it = s.caps->find(c->level);
-    if (it != s.caps->end()) {
-      at = std::min(at, it->second);
-      af = std::min(af, it->second);
+  if (inWorld) {
+    PhysicalCell* ph = static_cast<VirtualCell*>(c)->phys;
+    if (ph == nullptr) {
+      // unbound: will map onto SOME physical cell of this level — within
+      // the nearest bound ancestor's region if one exists (its physical
+      // identity pins the candidates), else anywhere in the chain
+      PhysicalCell* region = nullptr;
+      for (Cell* a = c->parent; a != nullptr; a = a->parent) {
+        PhysicalCell* ap = static_cast<VirtualCell*>(a)->phys;
+        if (ap != nullptr) {
+          region = ap;
+          break;
+        }
+      }
+      int cap = INT_MAX;
+      if (region != nullptr) {
+        cap = maxInWorldAtLevelUnder(region, c->level, *s.world);
+      } else {
+        auto it = s.world->caps.find(c->level);
+        if (it != s.world->caps.end()) cap = it->second;
+      }
+      if (cap != INT_MAX) {
+        at = std::min(at, cap);
+        af = std::min(af, cap);
+      }
+    } else {
+      // bound: the physical region is known; its in-world free capacity
+      // bounds what the subtree (incl. positionally-ambiguous unbound
+      // leaves) can deliver cleanly
+      int cap = physFreeInWorld(ph, *s.world);
+      // preemptible leaves are outside the free-world analysis; only the
+      // free component is capped, the preemptible surplus rides on top
+      int preemptible = at - af;
+      af = std::min(af, cap);
+      at = af + preemptible;
     }
   }
   return {at, af};
@@ -355,17 +408,42 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 
 // Max link-clean FREE leaf capacity within one cell: free healthy leaves
 // minus the minimum exclusions needed so no degraded link has both
-// endpoints in the set. Used to build SchedulingRequest::physCleanCaps.
+// endpoints in the set.
 int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes) {
   PickSession empty;
   empty.suggestedFilter = suggestedNodes;
   return cleanAvailAnalysis(c, kOpportunisticPriority, empty, INT_MAX, nullptr);
 }
 
+CleanShapeWorld computeCleanShapeWorld(const ChainCellList& ccl,
+                                       const std::set<std::string>* suggestedNodes) {
+  CleanShapeWorld w;
+  PickSession session;
+  session.suggestedFilter = suggestedNodes;
+  int top = ccl.top();
+  for (Cell* c : ccl.at(top)) {
+    std::vector<Cell*> excl;
+    cleanAvailAnalysis(c, kOpportunisticPriority, session, 0, &excl);
+    for (Cell* e : excl) {
+      session.excluded.insert(e);
+      w.excluded.insert(static_cast<PhysicalCell*>(e));
+    }
+  }
+  for (int l = kLowestLevel; l <= top; l++) {
+    int best = 0;
+    for (Cell* c : ccl.at(l)) {
+      best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+    }
+    w.caps[l] = best;
+  }
+  return w;
+}
+
+
 bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums, int priority,
                                           const std::set<std::string>& suggestedNodes,
                                           bool ignoreSuggestedNodes, long long minHbmBytes,
-                                          bool honorLinks, const std::map<int, int>* physCleanCaps,
+                                          bool honorLinks, const CleanShapeWorld* cleanWorld,
                                           Placement<Cell>* out,
                                           std::string* failedReason) const {
   // Build and sort the cluster view: healthy > suggested > same-priority used
@@ -374,7 +452,7 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   cv.reserve(viewCells_.size());
   PickSession probe;  // empty: availability before any placement
   probe.minHbm = minHbmBytes;
-  if (honorLinks) probe.caps = physCleanCaps;
+  if (honorLinks) probe.world = cleanWorld;
   for (Cell* c : viewCells_) {
     NodeView n;
     n.c = c;
@@ -391,9 +469,11 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
     // nodes stay usable for their healthy pairs/quads. In link-honoring mode
     // a node's capacity is its CLEAN capacity (largest co-placeable set with
     // no degraded xGMI link inside); the dirty retry lifts that.
-    if (honorLinks && badLinksUnderOf(c) > 0) {
+    if (honorLinks && probe.world == nullptr && badLinksUnderOf(c) > 0) {
+      // physical/bound view without a precomputed world: per-node analysis
       n.freeAtPriority = cleanAvailAnalysis(c, priority, probe, INT_MAX, nullptr);
     } else {
+      // with a world, availLeaves is already the in-world (clean) capacity
       n.freeAtPriority = availLeaves(c, priority, probe).first;
     }
     auto [healthy, suggested] = healthyAndSuggested(c, suggestedNodes, ignoreSuggestedNodes);
@@ -462,16 +542,20 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
   PickSession session;
   session.minHbm = minHbmBytes;
   if (honorLinks) {
-    session.caps = physCleanCaps;
+    session.world = cleanWorld;
     // Gang-wide link cleanliness per node: every pod of the gang on one node
     // communicates with every other (the gang's collective), so exclusions
     // are computed from the node's TOTAL gang demand before any pick.
-    std::map<int, int> nodeDemand;  // nodeIndex -> total gang leaves
-    for (size_t i = 0; i < sortedLeafNums.size(); i++) {
-      nodeDemand[pickedNodeIndices[i]] += sortedLeafNums[i];
-    }
-    for (auto& [nodeIndex, demand] : nodeDemand) {
-      if (demand >= 2) applyLinkExclusions(cv[nodeIndex].c, demand, priority, session);
+    // With a precomputed clean-shape world the exclusions are already
+    // global (world.excluded), so the per-node analysis is skipped.
+    if (cleanWorld == nullptr) {
+      std::map<int, int> nodeDemand;  // nodeIndex -> total gang leaves
+      for (size_t i = 0; i < sortedLeafNums.size(); i++) {
+        nodeDemand[pickedNodeIndices[i]] += sortedLeafNums[i];
+      }
+      for (auto& [nodeIndex, demand] : nodeDemand) {
+        if (demand >= 2) applyLinkExclusions(cv[nodeIndex].c, demand, priority, session);
+      }
     }
   }
   out->clear();
@@ -490,15 +574,15 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
                              const std::set<std::string>& suggestedNodes,
                              bool ignoreSuggestedNodes, Placement<Cell>* out,
                              std::string* failedReason, long long minHbmBytes,
-                             const std::map<int, int>* physCleanCaps) const {
+                             const CleanShapeWorld* cleanWorld) const {
   std::vector<int> sortedLeafNums;
   for (auto& [leafNum, podNum] : podLeafCellNums) {
     for (int i = 0; i < podNum; i++) sortedLeafNums.push_back(leafNum);
   }
-  // caps are only passed when the chain carries degraded links, so their
-  // presence alone demands the dirty-retry rungs of the ladder (virtual
-  // views are mostly unbound and show badLinksUnderOf == 0)
-  bool anyBadLinks = physCleanCaps != nullptr;
+  // a clean-shape world is only passed when the chain carries degraded
+  // links, so its presence alone demands the dirty-retry rungs of the
+  // ladder (virtual views are mostly unbound and show badLinksUnderOf == 0)
+  bool anyBadLinks = cleanWorld != nullptr;
   for (Cell* c : viewCells_) {
     if (anyBadLinks || badLinksUnderOf(c) > 0) {
       anyBadLinks = true;
@@ -511,7 +595,7 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
   // degraded link (slow xGMI), hence free+dirty before preempt+clean.
   auto attempt = [&](int p, bool honorLinks) {
     return tryScheduleAtPriority(sortedLeafNums, p, suggestedNodes, ignoreSuggestedNodes,
-                                 minHbmBytes, honorLinks, physCleanCaps, out, failedReason);
+                                 minHbmBytes, honorLinks, cleanWorld, out, failedReason);
   };
   if (attempt(kOpportunisticPriority, true)) return true;
   if (anyBadLinks && attempt(kOpportunisticPriority, false)) return true;
@@ -536,7 +620,7 @@ bool IntraVCScheduler::schedule(const SchedulingRequest& sr, Placement<VirtualCe
   if (scheduler == nullptr ||
       !scheduler->Schedule(sr.podLeafCellNums, sr.priority, *sr.suggestedNodes,
                            sr.ignoreSuggestedNodes, &generic, failedReason, sr.hbmBytes,
-                           sr.physCleanCaps)) {
+                           sr.cleanWorld)) {
     if (failedReason->empty()) *failedReason = "no scheduler for request";
     *failedReason += " when scheduling in VC " + sr.vc;
     return false;
