@@ -411,150 +411,152 @@ bool HivedCore::handleSchedulingRequest(const SchedulingRequest& sr, Placement<P
 
 bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement<PhysicalCell>* phys,
                                         Placement<VirtualCell>* virt, std::string* failedReason) {
-  // When the chain carries degraded xGMI links, feed the virtual descent the
-  // physical clean-shape caps (level -> max link-clean free capacity of any
-  // one physical cell) so its link-honoring attempts pick shapes that admit
-  // a clean mapping; the dirty rungs of the attempt ladder run uncapped.
   SchedulingRequest sr = srIn;
-  CleanShapeWorld cleanWorld;
-  {
-    auto& ccl = fullCellList_[sr.chain];
-    bool chainHasBadLinks = false;
-    for (Cell* c : ccl.at(ccl.top())) {
-      if (static_cast<PhysicalCell*>(c)->badLinksUnder > 0) {
-        chainHasBadLinks = true;
-        break;
-      }
-    }
-    int gangLeaves = 0;
-    for (auto& [ln, pn] : sr.podLeafCellNums) gangLeaves += ln * pn;
-    // a 1-leaf request never straddles a link: no world (it may freely use
-    // — and by the parking preference, should use — degraded endpoints)
-    if (chainHasBadLinks && gangLeaves >= 2) {
-      // honor the suggested-node restriction the mapping stage will apply:
-      // capacity on non-suggested nodes cannot satisfy this request
-      const std::set<std::string>* filter =
-          sr.ignoreSuggestedNodes ? nullptr : sr.suggestedNodes;
-      cleanWorld = computeCleanShapeWorld(ccl, filter);
-      sr.cleanWorld = &cleanWorld;
-      if (mapDebugRelease()) {
-        fprintf(stderr, "[world] chain=%s ignoreSug=%d excl=%zu:", sr.chain.c_str(),
-                (int)sr.ignoreSuggestedNodes, cleanWorld.excluded.size());
-        for (auto& [l, v] : cleanWorld.caps) fprintf(stderr, " L%d=%d", l, v);
-        fprintf(stderr, "\n");
-      }
-    }
-  }
-  if (!vcSchedulers_[sr.vc].schedule(sr, virt, failedReason)) return false;
-  if (mapDebugRelease()) {
-    for (auto& [ln, pods] : *virt) {
-      for (auto& pod : pods) {
-        fprintf(stderr, "[virt ln=%d]", ln);
-        for (auto* v : pod) fprintf(stderr, " %s", v->address.c_str());
-        fprintf(stderr, "\n");
-      }
-    }
-  }
 
-  std::unordered_map<VirtualCell*, PhysicalCell*> bindings;
-  auto lazyPreempted = tryLazyPreempt(*virt, sr.groupName);
-
-  // build binding paths: trees of unbound virtual cells to bind
-  std::vector<std::unique_ptr<BindingVertex>> roots;
-  std::vector<BindingVertex*> preassigned;
-  std::vector<std::vector<BindingVertex*>> nonPreassigned;
-  std::unordered_map<VirtualCell*, BindingVertex*> vertexMap;
-  for (auto& [leafNum, pods] : *virt) {
-    (void)leafNum;
-    for (auto& pod : pods) {
-      for (VirtualCell* leaf : pod) {
-        if (leaf->phys != nullptr) {
-          bindings[leaf] = leaf->phys;
-          continue;
-        }
-        std::vector<VirtualCell*> path;
-        for (Cell* c = leaf; c != nullptr; c = c->parent) {
-          auto* vc = static_cast<VirtualCell*>(c);
-          if (vc->phys != nullptr || vertexMap.count(vc)) break;
-          path.push_back(vc);
-        }
-        if (path.empty()) continue;
-        VirtualCell* pathRoot = path.back();
-        auto rootVertex = std::make_unique<BindingVertex>();
-        rootVertex->cell = pathRoot;
-        BindingVertex* rootPtr = rootVertex.get();
-        vertexMap[pathRoot] = rootPtr;
-        if (pathRoot->parent == nullptr) {
-          preassigned.push_back(rootPtr);
-          roots.push_back(std::move(rootVertex));
-        } else if (static_cast<VirtualCell*>(pathRoot->parent)->phys != nullptr) {
-          bool buddyExists = false;
-          for (auto& grp : nonPreassigned) {
-            if (grp[0]->cell->parent == pathRoot->parent) {
-              grp.push_back(rootPtr);
-              buddyExists = true;
-              break;
-            }
-          }
-          if (!buddyExists) nonPreassigned.push_back({rootPtr});
-          roots.push_back(std::move(rootVertex));
-        } else {
-          BindingVertex* parentVertex = vertexMap.at(static_cast<VirtualCell*>(pathRoot->parent));
-          parentVertex->children.push_back(std::move(rootVertex));
-        }
-        for (int i = static_cast<int>(path.size()) - 2; i >= 0; i--) {
-          auto childVertex = std::make_unique<BindingVertex>();
-          childVertex->cell = path[i];
-          vertexMap[path[i]] = childVertex.get();
-          vertexMap.at(static_cast<VirtualCell*>(path[i]->parent))
-              ->children.push_back(std::move(childVertex));
+  // One full placement attempt: intra-VC schedule -> lazy preemption ->
+  // binding-vertex construction -> virtual->physical mapping. honorOnly
+  // restricts both the descent and the mapping to link-clean choices
+  // within sr.cleanWorld; reverts lazy preemption on failure.
+  auto attemptOnce = [&](bool honorOnly) -> bool {
+    sr.honorLinksOnly = honorOnly;
+    if (!vcSchedulers_[sr.vc].schedule(sr, virt, failedReason)) return false;
+    if (mapDebugRelease()) {
+      for (auto& [ln, pods] : *virt) {
+        for (auto& pod : pods) {
+          fprintf(stderr, "[virt ln=%d honorOnly=%d]", ln, (int)honorOnly);
+          for (auto* v : pod) fprintf(stderr, " %s", v->address.c_str());
+          fprintf(stderr, "\n");
         }
       }
     }
-  }
 
-  // Two-tier mapping when the chain carries degraded xGMI links: first
-  // demand a link-clean mapping (no two gang leaves across a degraded link;
-  // backtracking explores alternatives), then accept a dirty one — capacity
-  // guarantees outrank link quality. Each attempt gets fresh free-list /
-  // accounting copies (the mapper mutates them).
-  bool anyBadLinks = false;
-  {
-    auto& ccl = fullCellList_[sr.chain];
-    for (Cell* c : ccl.at(ccl.top())) {
-      if (static_cast<PhysicalCell*>(c)->badLinksUnder > 0) {
-        anyBadLinks = true;
-        break;
-      }
-    }
-  }
-  const std::unordered_map<VirtualCell*, PhysicalCell*> seedBindings = bindings;
-  auto tryMap = [&](bool honorLinks) {
-    bindings = seedBindings;  // drop partial picks from a failed attempt
-    std::map<int, int> freeCellNumCopy = allVCFreeCellNum_[sr.chain];
-    return mapVirtualPlacementToPhysical(preassigned, nonPreassigned,
-                                         freeCellList_[sr.chain].shallowCopy(), freeCellNumCopy,
-                                         *sr.suggestedNodes, sr.ignoreSuggestedNodes, bindings,
-                                         sr.hbmBytes, honorLinks, sr.cleanWorld);
-  };
-  bool mapped = anyBadLinks ? (tryMap(true) || tryMap(false)) : tryMap(false);
-  if (mapped) {
-    phys->clear();
+    std::unordered_map<VirtualCell*, PhysicalCell*> bindings;
+    auto lazyPreempted = tryLazyPreempt(*virt, sr.groupName);
+
+    // build binding paths: trees of unbound virtual cells to bind
+    std::vector<std::unique_ptr<BindingVertex>> roots;
+    std::vector<BindingVertex*> preassigned;
+    std::vector<std::vector<BindingVertex*>> nonPreassigned;
+    std::unordered_map<VirtualCell*, BindingVertex*> vertexMap;
     for (auto& [leafNum, pods] : *virt) {
+      (void)leafNum;
       for (auto& pod : pods) {
-        std::vector<PhysicalCell*> cells;
-        cells.reserve(pod.size());
-        for (VirtualCell* v : pod) cells.push_back(bindings.at(v));
-        (*phys)[leafNum].push_back(std::move(cells));
+        for (VirtualCell* leaf : pod) {
+          if (leaf->phys != nullptr) {
+            bindings[leaf] = leaf->phys;
+            continue;
+          }
+          std::vector<VirtualCell*> path;
+          for (Cell* c = leaf; c != nullptr; c = c->parent) {
+            auto* vc = static_cast<VirtualCell*>(c);
+            if (vc->phys != nullptr || vertexMap.count(vc)) break;
+            path.push_back(vc);
+          }
+          if (path.empty()) continue;
+          VirtualCell* pathRoot = path.back();
+          auto rootVertex = std::make_unique<BindingVertex>();
+          rootVertex->cell = pathRoot;
+          BindingVertex* rootPtr = rootVertex.get();
+          vertexMap[pathRoot] = rootPtr;
+          if (pathRoot->parent == nullptr) {
+            preassigned.push_back(rootPtr);
+            roots.push_back(std::move(rootVertex));
+          } else if (static_cast<VirtualCell*>(pathRoot->parent)->phys != nullptr) {
+            bool buddyExists = false;
+            for (auto& grp : nonPreassigned) {
+              if (grp[0]->cell->parent == pathRoot->parent) {
+                grp.push_back(rootPtr);
+                buddyExists = true;
+                break;
+              }
+            }
+            if (!buddyExists) nonPreassigned.push_back({rootPtr});
+            roots.push_back(std::move(rootVertex));
+          } else {
+            BindingVertex* parentVertex = vertexMap.at(static_cast<VirtualCell*>(pathRoot->parent));
+            parentVertex->children.push_back(std::move(rootVertex));
+          }
+          for (int i = static_cast<int>(path.size()) - 2; i >= 0; i--) {
+            auto childVertex = std::make_unique<BindingVertex>();
+            childVertex->cell = path[i];
+            vertexMap[path[i]] = childVertex.get();
+            vertexMap.at(static_cast<VirtualCell*>(path[i]->parent))
+                ->children.push_back(std::move(childVertex));
+          }
+        }
       }
     }
-    return true;
+
+    const std::unordered_map<VirtualCell*, PhysicalCell*> seedBindings = bindings;
+    auto tryMap = [&](bool honorLinks) {
+      bindings = seedBindings;  // drop partial picks from a failed attempt
+      std::map<int, int> freeCellNumCopy = allVCFreeCellNum_[sr.chain];
+      return mapVirtualPlacementToPhysical(preassigned, nonPreassigned,
+                                           freeCellList_[sr.chain].shallowCopy(), freeCellNumCopy,
+                                           *sr.suggestedNodes, sr.ignoreSuggestedNodes, bindings,
+                                           sr.hbmBytes, honorLinks, sr.cleanWorld);
+    };
+    bool mapped;
+    if (honorOnly) {
+      mapped = tryMap(true);
+    } else if (sr.cleanWorld != nullptr || chainHasBadLinks(sr.chain)) {
+      mapped = tryMap(true) || tryMap(false);
+    } else {
+      mapped = tryMap(false);
+    }
+    if (mapped) {
+      phys->clear();
+      for (auto& [leafNum, pods] : *virt) {
+        for (auto& pod : pods) {
+          std::vector<PhysicalCell*> cells;
+          cells.reserve(pod.size());
+          for (VirtualCell* v : pod) cells.push_back(bindings.at(v));
+          (*phys)[leafNum].push_back(std::move(cells));
+        }
+      }
+      return true;
+    }
+    for (auto& [groupName, placement] : lazyPreempted) {
+      revertLazyPreempt(groups_.at(groupName).get(), placement);
+    }
+    return false;
+  };
+
+  // When the chain carries degraded xGMI links and the gang needs >= 2
+  // leaves, try every enumerated clean-shape world (different consistent
+  // endpoint-avoidance choices admit different clean shapes) with
+  // link-honoring-only rungs, then fall back to the unconstrained ladder
+  // (capacity outranks link quality). A 1-leaf request never straddles a
+  // link: no worlds (it may freely use — and by the parking preference,
+  // should use — degraded endpoints).
+  int gangLeaves = 0;
+  for (auto& [ln, pn] : sr.podLeafCellNums) gangLeaves += ln * pn;
+  if (chainHasBadLinks(sr.chain) && gangLeaves >= 2) {
+    const std::set<std::string>* filter =
+        sr.ignoreSuggestedNodes ? nullptr : sr.suggestedNodes;
+    auto worlds = computeCleanShapeWorlds(fullCellList_[sr.chain], filter, 4);
+    for (auto& w : worlds) {
+      sr.cleanWorld = &w;
+      if (mapDebugRelease()) {
+        fprintf(stderr, "[world] chain=%s excl=%zu:", sr.chain.c_str(), w.excluded.size());
+        for (auto& [l, v] : w.caps) fprintf(stderr, " L%d=%d", l, v);
+        fprintf(stderr, "\n");
+      }
+      if (attemptOnce(true)) return true;
+    }
+    sr.cleanWorld = nullptr;
   }
-  for (auto& [groupName, placement] : lazyPreempted) {
-    revertLazyPreempt(groups_.at(groupName).get(), placement);
-  }
+  if (attemptOnce(false)) return true;
   *failedReason = std::string("Mapping the virtual placement would need to use at least one ") +
                   (sr.ignoreSuggestedNodes ? "bad" : "bad or non-suggested") + " node";
+  return false;
+}
+
+bool HivedCore::chainHasBadLinks(const std::string& chain) {
+  auto& ccl = fullCellList_[chain];
+  for (Cell* c : ccl.at(ccl.top())) {
+    if (static_cast<PhysicalCell*>(c)->badLinksUnder > 0) return true;
+  }
   return false;
 }
 

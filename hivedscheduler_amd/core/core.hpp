@@ -282,6 +282,9 @@ using Placement = std::map<int, std::vector<std::vector<CellT*>>>;
 struct CleanShapeWorld {
   std::map<int, int> caps;
   std::unordered_set<PhysicalCell*> excluded;
+  // unbound physical cells per level, for hint resolution when no bound
+  // ancestor scopes the candidates (see topo_sched.cpp pickLeavesWorld)
+  std::map<int, std::vector<PhysicalCell*>> physByLevel;
 };
 
 struct SchedulingRequest {
@@ -304,6 +307,9 @@ struct SchedulingRequest {
   // shape up front instead of falling to a dirty placement. Dirty rungs
   // run without the world, so capacity is never sacrificed.
   const CleanShapeWorld* cleanWorld = nullptr;
+  // run only the link-honoring placement rungs (the caller iterates over
+  // several clean-shape worlds and provides the dirty fallback itself)
+  bool honorLinksOnly = false;
 };
 
 // ---------------------------------------------------------------------------
@@ -324,7 +330,7 @@ class TopoScheduler {
                 const std::set<std::string>& suggestedNodes, bool ignoreSuggestedNodes,
                 Placement<Cell>* out, std::string* failedReason,
                 long long minHbmBytes = 0,
-                const CleanShapeWorld* cleanWorld = nullptr) const;
+                const CleanShapeWorld* cleanWorld = nullptr, bool honorOnly = false) const;
 
  private:
   struct NodeView {
@@ -483,6 +489,7 @@ class HivedCore {
                                std::string* failedReason);
   bool scheduleGuaranteedGroup(const SchedulingRequest& sr, Placement<PhysicalCell>* phys,
                                Placement<VirtualCell>* virt, std::string* failedReason);
+  bool chainHasBadLinks(const std::string& chain);
   bool scheduleOpportunisticGroup(const SchedulingRequest& sr, Placement<PhysicalCell>* phys,
                                   std::string* failedReason);
   void validateSchedulingRequest(const SchedulingRequest& sr, const std::string& podKey);
@@ -584,9 +591,11 @@ Cell* ancestorNoHigherThanNode(Cell* c);
 // max link-clean FREE leaf capacity within one cell (no session state);
 // suggestedNodes non-null restricts usable leaves to those nodes
 int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes = nullptr);
-// the chain's clean-shape world under one consistent endpoint choice
-CleanShapeWorld computeCleanShapeWorld(const ChainCellList& ccl,
-                                       const std::set<std::string>* suggestedNodes);
+// the chain's clean-shape worlds: one per enumerated consistent endpoint
+// choice (different max independent sets admit different clean shapes)
+std::vector<CleanShapeWorld> computeCleanShapeWorlds(
+    const ChainCellList& ccl, const std::set<std::string>* suggestedNodes,
+    size_t maxWorlds = 4);
 void checkInvariants(const HivedCore& core);
 
 }  // namespace hived

@@ -17,6 +17,7 @@
 //  - per-leaf HBM capacity: a request may demand hbmBytesPerCell; leaves
 //    whose measured capacity falls short are unavailable for it.
 #include <climits>
+#include <functional>
 
 #include "core.hpp"
 
@@ -242,35 +243,35 @@ unsigned long long maxIndepSet(const std::vector<unsigned long long>& adj,
 // max clean count; when `excludeOut` is non-null and the clean capacity
 // covers `need`, fills it with the leaves to exclude (the bad-link-incident
 // leaves NOT in the chosen independent set).
-int cleanAvailAnalysis(Cell* viewCell, int p, const PickSession& s, int need,
-                       std::vector<Cell*>* excludeOut) {
-  auto [availTotal, availFree] = availLeaves(viewCell, p, s);
-  (void)availFree;
-  if (badLinksUnderOf(viewCell) == 0) return availTotal;
+// The bad-link graph over a view's available leaves: vertex v = incident
+// available leaf (index into `leaves` via verts[v]); adjacency as bitmasks.
+struct BadLinkGraph {
   std::vector<Cell*> leaves;
-  collectAvailableLeaves(viewCell, p, s, leaves);
-  // map physical leaf -> view leaf to resolve link endpoints in this view
-  std::unordered_map<PhysicalCell*, int> physIdx;
-  for (size_t i = 0; i < leaves.size(); i++) {
-    PhysicalCell* ph = physOf(leaves[i]);
-    if (ph != nullptr) physIdx[ph] = static_cast<int>(i);
-  }
-  // vertices = available leaves incident to a bad link whose peer is also
-  // available in this view
   std::vector<int> verts;
   std::vector<unsigned long long> adj;
+};
+
+BadLinkGraph buildBadLinkGraph(Cell* viewCell, int p, const PickSession& s) {
+  BadLinkGraph g;
+  collectAvailableLeaves(viewCell, p, s, g.leaves);
+  // map physical leaf -> view leaf to resolve link endpoints in this view
+  std::unordered_map<PhysicalCell*, int> physIdx;
+  for (size_t i = 0; i < g.leaves.size(); i++) {
+    PhysicalCell* ph = physOf(g.leaves[i]);
+    if (ph != nullptr) physIdx[ph] = static_cast<int>(i);
+  }
   std::unordered_map<int, int> leafToVert;
   auto vertOf = [&](int leafIdx) {
     auto it = leafToVert.find(leafIdx);
     if (it != leafToVert.end()) return it->second;
-    int v = static_cast<int>(verts.size());
+    int v = static_cast<int>(g.verts.size());
     leafToVert[leafIdx] = v;
-    verts.push_back(leafIdx);
-    adj.push_back(0);
+    g.verts.push_back(leafIdx);
+    g.adj.push_back(0);
     return v;
   };
-  for (size_t i = 0; i < leaves.size(); i++) {
-    PhysicalCell* ph = physOf(leaves[i]);
+  for (size_t i = 0; i < g.leaves.size(); i++) {
+    PhysicalCell* ph = physOf(g.leaves[i]);
     if (ph == nullptr || ph->badLinkPeers.empty()) continue;
     for (PhysicalCell* peer : ph->badLinkPeers) {
       auto it = physIdx.find(peer);
@@ -278,23 +279,83 @@ int cleanAvailAnalysis(Cell* viewCell, int p, const PickSession& s, int need,
       int a = vertOf(static_cast<int>(i));
       int b = vertOf(it->second);
       if (a < 64 && b < 64) {
-        adj[a] |= 1ull << b;
-        adj[b] |= 1ull << a;
+        g.adj[a] |= 1ull << b;
+        g.adj[b] |= 1ull << a;
       }
     }
   }
-  int nv = static_cast<int>(verts.size());
+  return g;
+}
+
+int cleanAvailAnalysis(Cell* viewCell, int p, const PickSession& s, int need,
+                       std::vector<Cell*>* excludeOut) {
+  auto [availTotal, availFree] = availLeaves(viewCell, p, s);
+  (void)availFree;
+  if (badLinksUnderOf(viewCell) == 0) return availTotal;
+  BadLinkGraph g = buildBadLinkGraph(viewCell, p, s);
+  int nv = static_cast<int>(g.verts.size());
   if (nv == 0) return availTotal;
   if (nv > 60) return availTotal - nv;  // degenerate; be conservative
   unsigned long long all = nv == 64 ? ~0ull : ((1ull << nv) - 1);
-  unsigned long long chosen = maxIndepSet(adj, all);
+  unsigned long long chosen = maxIndepSet(g.adj, all);
   int clean = availTotal - nv + __builtin_popcountll(chosen);
   if (excludeOut != nullptr && clean >= need) {
     for (int v = 0; v < nv; v++) {
-      if (!(chosen & (1ull << v))) excludeOut->push_back(leaves[verts[v]]);
+      if (!(chosen & (1ull << v))) excludeOut->push_back(g.leaves[g.verts[v]]);
     }
   }
   return clean;
+}
+
+// All maximum independent sets of the graph (masks), deduped, capped.
+void enumerateMaxIndepMasks(const std::vector<unsigned long long>& adj,
+                            unsigned long long cand, unsigned long long acc,
+                            std::vector<unsigned long long>& out, size_t cap) {
+  if (cand == 0) {
+    out.push_back(acc);
+    return;
+  }
+  if (out.size() >= cap * 8) return;  // bound the search
+  int v = __builtin_ctzll(cand);
+  unsigned long long bit = 1ull << v;
+  enumerateMaxIndepMasks(adj, cand & ~bit & ~adj[v], acc | bit, out, cap);
+  enumerateMaxIndepMasks(adj, cand & ~bit, acc, out, cap);
+}
+
+// Exclusion-set VARIANTS for a top cell: every maximum independent set of
+// its bad-link graph yields one valid "which endpoints to avoid" choice;
+// different choices admit different clean shapes, so the ladder tries a few.
+std::vector<std::vector<Cell*>> enumerateCleanExclusionVariants(Cell* top,
+                                                                const PickSession& s,
+                                                                size_t cap) {
+  std::vector<std::vector<Cell*>> out;
+  if (badLinksUnderOf(top) == 0) return out;
+  BadLinkGraph g = buildBadLinkGraph(top, kOpportunisticPriority, s);
+  int nv = static_cast<int>(g.verts.size());
+  if (nv == 0 || nv > 24) {
+    if (nv > 24) {  // degenerate: one conservative variant via the analysis
+      std::vector<Cell*> excl;
+      cleanAvailAnalysis(top, kOpportunisticPriority, s, 0, &excl);
+      if (!excl.empty()) out.push_back(std::move(excl));
+    }
+    return out;
+  }
+  unsigned long long all = nv == 64 ? ~0ull : ((1ull << nv) - 1);
+  std::vector<unsigned long long> masks;
+  enumerateMaxIndepMasks(g.adj, all, 0, masks, cap);
+  size_t best = 0;
+  for (auto m : masks) best = std::max(best, (size_t)__builtin_popcountll(m));
+  std::set<unsigned long long> seen;
+  for (auto m : masks) {
+    if ((size_t)__builtin_popcountll(m) != best || !seen.insert(m).second) continue;
+    std::vector<Cell*> excl;
+    for (int v = 0; v < nv; v++) {
+      if (!(m & (1ull << v))) excl.push_back(g.leaves[g.verts[v]]);
+    }
+    out.push_back(std::move(excl));
+    if (out.size() >= cap) break;
+  }
+  return out;
 }
 
 // Exclude, for the rest of this session (= this gang), the bad-link-incident
@@ -395,6 +456,147 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
   if (remaining != 0) throw HivedError::Internal("pickLeaves underflow in " + cell->address);
 }
 
+// World-mode descent with PHYSICAL MIRRORING: each virtual cell is picked
+// against a tentative physical "hint" cell whose in-world structure the
+// drain mirrors exactly. This is what makes multi-link shapes exact: level-
+// max caps cannot express "this quad's two in-world leaves sit in DIFFERENT
+// pairs", but the hint's own children can. Bound virtual children identify
+// their hint child directly; unbound ones (all-unbound subtrees, since
+// bindings are upward-contiguous) are rank-matched to the hint's unbound
+// children by capacity. Returns false (instead of throwing) when the shape
+// cannot be realized — the caller falls to the next world / dirty rung.
+bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& s,
+                     std::vector<Cell*>& out) {
+  if (cell->level == kLowestLevel) {
+    // re-check eligibility (the caller's counts came from caps, not hints)
+    auto [t, f] = availLeaves(cell, p, s);
+    (void)f;
+    if (t < 1) return false;
+    s.taken.insert(cell);
+    out.push_back(cell);
+    return true;
+  }
+  struct Pair {
+    Cell* child;
+    PhysicalCell* hint;
+    int avail;
+  };
+  std::vector<Pair> pairs;
+  std::vector<Cell*> unboundChildren;
+  std::vector<PhysicalCell*> freeHints;
+  for (Cell* childC : cell->children) {
+    auto* vc = static_cast<VirtualCell*>(childC);
+    if (vc->phys != nullptr) {
+      pairs.push_back({vc, vc->phys, 0});
+    } else {
+      unboundChildren.push_back(vc);
+    }
+  }
+  for (Cell* hc : hint->children) {
+    auto* ph = static_cast<PhysicalCell*>(hc);
+    if (ph->virt == nullptr) freeHints.push_back(ph);
+  }
+  if (freeHints.size() < unboundChildren.size()) return false;
+  // rank-match: unbound virtual children are interchangeable, so order by
+  // the hint children's in-world capacity descending (tightest structure
+  // last). Sort virtual children by raw availability desc to pair the most
+  // capable with the most capable.
+  std::stable_sort(freeHints.begin(), freeHints.end(), [&](PhysicalCell* a, PhysicalCell* b) {
+    return physFreeInWorld(a, *s.world) > physFreeInWorld(b, *s.world);
+  });
+  std::stable_sort(unboundChildren.begin(), unboundChildren.end(), [&](Cell* a, Cell* b) {
+    return availLeaves(a, p, s).first > availLeaves(b, p, s).first;
+  });
+  for (size_t i = 0; i < unboundChildren.size(); i++) {
+    pairs.push_back({unboundChildren[i], freeHints[i], 0});
+  }
+  int total = 0;
+  for (auto& pr : pairs) {
+    pr.avail = std::min(availLeaves(pr.child, p, s).first, physFreeInWorld(pr.hint, *s.world));
+    total += pr.avail;
+  }
+  if (total < q) return false;
+  // single child fits: tightest such pair (packing); LCA-minimal since the
+  // recursion descends whenever possible
+  int best = -1;
+  for (size_t i = 0; i < pairs.size(); i++) {
+    if (pairs[i].avail >= q && (best < 0 || pairs[i].avail < pairs[best].avail)) {
+      best = static_cast<int>(i);
+    }
+  }
+  if (best >= 0) {
+    return pickLeavesWorld(pairs[best].child, pairs[best].hint, q, p, s, out);
+  }
+  // drain, largest first
+  std::stable_sort(pairs.begin(), pairs.end(), [](const Pair& a, const Pair& b) {
+    return a.avail > b.avail;
+  });
+  int remaining = q;
+  size_t outStart = out.size();
+  for (auto& pr : pairs) {
+    if (remaining == 0) break;
+    int t = std::min(remaining, pr.avail);
+    if (t > 0) {
+      if (!pickLeavesWorld(pr.child, pr.hint, t, p, s, out)) {
+        // roll back this request's takes and report shape failure
+        for (size_t i = outStart; i < out.size(); i++) s.taken.erase(out[i]);
+        out.resize(outStart);
+        return false;
+      }
+      remaining -= t;
+    }
+  }
+  if (remaining != 0) {
+    for (size_t i = outStart; i < out.size(); i++) s.taken.erase(out[i]);
+    out.resize(outStart);
+    return false;
+  }
+  return true;
+}
+
+// Resolve the tentative physical target for a world-mode pick at `cell`:
+// its own binding if bound; otherwise the tightest-fit unbound physical
+// cell with in-world capacity >= q — scoped under the nearest bound
+// ancestor's region, or chain-wide via the world's per-level cell list.
+PhysicalCell* resolveWorldHint(Cell* cell, int q, const PickSession& s) {
+  auto* vc = static_cast<VirtualCell*>(cell);
+  if (vc->phys != nullptr) return vc->phys;
+  PhysicalCell* region = nullptr;
+  for (Cell* a = cell->parent; a != nullptr; a = a->parent) {
+    PhysicalCell* ap = static_cast<VirtualCell*>(a)->phys;
+    if (ap != nullptr) {
+      region = ap;
+      break;
+    }
+  }
+  PhysicalCell* best = nullptr;
+  int bestCap = INT_MAX;
+  auto consider = [&](PhysicalCell* ph) {
+    if (ph->virt != nullptr) return;
+    int cap = physFreeInWorld(ph, *s.world);
+    if (cap >= q && cap < bestCap) {
+      best = ph;
+      bestCap = cap;
+    }
+  };
+  if (region != nullptr) {
+    std::function<void(PhysicalCell*)> walk = [&](PhysicalCell* ph) {
+      if (ph->level == cell->level) {
+        consider(ph);
+        return;
+      }
+      for (Cell* c : ph->children) walk(static_cast<PhysicalCell*>(c));
+    };
+    walk(region);
+  } else {
+    auto it = s.world->physByLevel.find(cell->level);
+    if (it != s.world->physByLevel.end()) {
+      for (PhysicalCell* ph : it->second) consider(ph);
+    }
+  }
+  return best;
+}
+
 std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>& suggestedNodes,
                                            bool ignoreSuggestedNodes) {
   PhysicalCell* pc = physOf(c);
@@ -415,28 +617,56 @@ int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes) 
   return cleanAvailAnalysis(c, kOpportunisticPriority, empty, INT_MAX, nullptr);
 }
 
-CleanShapeWorld computeCleanShapeWorld(const ChainCellList& ccl,
-                                       const std::set<std::string>* suggestedNodes) {
-  CleanShapeWorld w;
-  PickSession session;
-  session.suggestedFilter = suggestedNodes;
+std::vector<CleanShapeWorld> computeCleanShapeWorlds(
+    const ChainCellList& ccl, const std::set<std::string>* suggestedNodes, size_t maxWorlds) {
+  PickSession base;
+  base.suggestedFilter = suggestedNodes;
   int top = ccl.top();
+  // per top cell: the exclusion-set variants (every max independent set of
+  // its bad-link graph). Links are node-local, so top cells are independent.
+  std::vector<std::vector<std::vector<Cell*>>> perCell;
   for (Cell* c : ccl.at(top)) {
-    std::vector<Cell*> excl;
-    cleanAvailAnalysis(c, kOpportunisticPriority, session, 0, &excl);
-    for (Cell* e : excl) {
-      session.excluded.insert(e);
-      w.excluded.insert(static_cast<PhysicalCell*>(e));
-    }
+    auto variants = enumerateCleanExclusionVariants(c, base, maxWorlds);
+    if (!variants.empty()) perCell.push_back(std::move(variants));
   }
-  for (int l = kLowestLevel; l <= top; l++) {
-    int best = 0;
-    for (Cell* c : ccl.at(l)) {
-      best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+  std::vector<CleanShapeWorld> worlds;
+  // odometer over the per-cell variants, capped
+  std::vector<size_t> idx(perCell.size(), 0);
+  for (;;) {
+    CleanShapeWorld w;
+    PickSession session = base;
+    for (size_t i = 0; i < perCell.size(); i++) {
+      for (Cell* e : perCell[i][idx[i]]) {
+        session.excluded.insert(e);
+        w.excluded.insert(static_cast<PhysicalCell*>(e));
+      }
     }
-    w.caps[l] = best;
+    for (int l = kLowestLevel; l <= top; l++) {
+      int best = 0;
+      for (Cell* c : ccl.at(l)) {
+        best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+      }
+      w.caps[l] = best;
+    }
+    for (int l = kLowestLevel; l <= top; l++) {
+      for (Cell* c : ccl.at(l)) {
+        auto* ph = static_cast<PhysicalCell*>(c);
+        if (ph->virt == nullptr) w.physByLevel[l].push_back(ph);
+      }
+    }
+    worlds.push_back(std::move(w));
+    if (worlds.size() >= maxWorlds || perCell.empty()) break;
+    // advance the odometer
+    size_t d = 0;
+    while (d < perCell.size()) {
+      if (++idx[d] < perCell[d].size()) break;
+      idx[d] = 0;
+      d++;
+    }
+    if (d == perCell.size()) break;
   }
-  return w;
+  if (worlds.empty()) worlds.push_back(CleanShapeWorld{});
+  return worlds;
 }
 
 
@@ -564,7 +794,17 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
     Cell* node = cv[pickedNodeIndices[i]].c;
     std::vector<Cell*> leaves;
     leaves.reserve(q);
-    pickLeaves(node, q, priority, session, leaves);
+    if (session.world != nullptr && !node->isPhysical()) {
+      // world mode on a virtual view: physical-mirroring descent; a shape
+      // that cannot be realized fails the attempt (next world / dirty rung)
+      PhysicalCell* hint = resolveWorldHint(node, q, session);
+      if (hint == nullptr || !pickLeavesWorld(node, hint, q, priority, session, leaves)) {
+        *failedReason = "no link-clean shape in this world";
+        return false;
+      }
+    } else {
+      pickLeaves(node, q, priority, session, leaves);
+    }
     (*out)[q].push_back(std::move(leaves));
   }
   return true;
@@ -574,7 +814,7 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
                              const std::set<std::string>& suggestedNodes,
                              bool ignoreSuggestedNodes, Placement<Cell>* out,
                              std::string* failedReason, long long minHbmBytes,
-                             const CleanShapeWorld* cleanWorld) const {
+                             const CleanShapeWorld* cleanWorld, bool honorOnly) const {
   std::vector<int> sortedLeafNums;
   for (auto& [leafNum, podNum] : podLeafCellNums) {
     for (int i = 0; i < podNum; i++) sortedLeafNums.push_back(leafNum);
@@ -593,11 +833,18 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
   // state (capacity outranks link quality), then the same two with
   // preemption. Avoiding a preemption (killing pods) outranks avoiding a
   // degraded link (slow xGMI), hence free+dirty before preempt+clean.
+  // honorOnly (set when the caller iterates over several clean-shape
+  // worlds) runs only the link-honoring rungs; the caller provides the
+  // dirty fallback itself after every world failed.
   auto attempt = [&](int p, bool honorLinks) {
     return tryScheduleAtPriority(sortedLeafNums, p, suggestedNodes, ignoreSuggestedNodes,
                                  minHbmBytes, honorLinks, cleanWorld, out, failedReason);
   };
   if (attempt(kOpportunisticPriority, true)) return true;
+  // world-mode (honorOnly) attempts run at the FREE tier only: the world is
+  // built from free leaves and its hint math does not model preemptible
+  // cells; preemptive clean placement falls to the no-world honor rung
+  if (honorOnly) return false;
   if (anyBadLinks && attempt(kOpportunisticPriority, false)) return true;
   if (priority > kOpportunisticPriority) {
     if (attempt(priority, true)) return true;
@@ -620,7 +867,7 @@ bool IntraVCScheduler::schedule(const SchedulingRequest& sr, Placement<VirtualCe
   if (scheduler == nullptr ||
       !scheduler->Schedule(sr.podLeafCellNums, sr.priority, *sr.suggestedNodes,
                            sr.ignoreSuggestedNodes, &generic, failedReason, sr.hbmBytes,
-                           sr.cleanWorld)) {
+                           sr.cleanWorld, sr.honorLinksOnly)) {
     if (failedReason->empty()) *failedReason = "no scheduler for request";
     *failedReason += " when scheduling in VC " + sr.vc;
     return false;

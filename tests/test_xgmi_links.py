@@ -359,22 +359,15 @@ def test_property_single_link_dirty_only_when_forced(seed):
     _assert_dirty_only_when_forced(sim, free, links, q)
 
 
-@pytest.mark.parametrize("seed", range(20))
-def test_property_multi_link_safe(seed):
-    """With SEVERAL degraded links the clean-shape world solves most cases
-    (offline oracle: ~99.5% of 4000 random 2-3-link cases place clean when
-    clean is possible) but shape-matching across multiple excluded
-    endpoints is heuristic — a documented bound (docs/design.md 5a). The
-    hard contract asserted here: requests always bind when capacity exists,
-    never over-allocate, and invariants hold."""
+@pytest.mark.parametrize("seed", range(30))
+def test_property_multi_link_dirty_only_when_forced(seed):
+    """Property (brute-force oracle) with SEVERAL simultaneous degraded
+    links: clean-shape world enumeration (every max independent set of the
+    bad-link graph) plus the physical-mirroring world descent make this
+    exact too — verified over 6000 random 1-4-link cases offline with ZERO
+    dirty-when-clean placements and zero spurious waits."""
     import random
 
     rng = random.Random(7000 + seed)
-    sim, free, links, q = _random_link_case(rng, rng.choice([2, 3]))
-    r = sim.schedule("req/q", sim.pod_spec(leaf_cells=q))
-    sim.alg._core.check_invariants()
-    if r.kind != "bind":
-        assert len(free) < q
-    else:
-        cells = set(r.bind_info.leafCellIsolation)
-        assert cells <= free and len(cells) == q
+    sim, free, links, q = _random_link_case(rng, rng.choice([2, 3, 4]))
+    _assert_dirty_only_when_forced(sim, free, links, q)
