@@ -51,7 +51,9 @@ class TraceReplay:
         self.decision_lat_ms = []
         self.queue_delay = []  # virtual-time arrival -> bind
         self.stats = {"binds": 0, "waits": 0, "preemptions": 0, "completions": 0,
-                      "victim_resubmits": 0, "invariant_checks": 0}
+                      "victim_resubmits": 0, "invariant_checks": 0,
+                      "link_degradations": 0, "link_heals": 0,
+                      "gpu_failures": 0, "gpu_recoveries": 0}
         self.invariant_every = invariant_every
         self.arrival_mean_s = arrival_mean_s
         self.decisions = 0
@@ -135,7 +137,7 @@ class TraceReplay:
         self.push(self.now + 5.0, "retry", job)  # default-scheduler retry cadence
         return False
 
-    def run(self, n_jobs):
+    def run(self, n_jobs, hardware_flaps=True):
         t = 0.0
         for i in range(n_jobs):
             t += self.rng.expovariate(1 / self.arrival_mean_s)
@@ -143,6 +145,25 @@ class TraceReplay:
             self.jobs[job["key"]] = job
             self.push(t, "arrive", job)
         horizon = t + 3600.0
+        if hardware_flaps:
+            # hardware weather riding over the whole trace: xGMI links
+            # degrade (agent pair probe / p2p matrix verdicts) and heal,
+            # GPUs fail and recover — exercising the first-class link
+            # machinery and leaf-health paths under churn
+            ft = 0.0
+            while ft < t:
+                ft += self.rng.expovariate(1 / 900.0)
+                node = f"node{self.rng.randrange(self.nodes) + 1}"
+                a, b = self.rng.sample(range(8), 2)
+                self.push(ft, "linkflap", (node, a, b))
+                self.push(ft + self.rng.expovariate(1 / 600.0), "linkheal", (node, a, b))
+            ft = 0.0
+            while ft < t:
+                ft += self.rng.expovariate(1 / 1800.0)
+                node = f"node{self.rng.randrange(self.nodes) + 1}"
+                g = self.rng.randrange(8)
+                self.push(ft, "gpufail", (node, g))
+                self.push(ft + self.rng.expovariate(1 / 900.0), "gpuheal", (node, g))
         check = self.sim.alg._core.check_invariants
         while self.events:
             self.now, _, kind, job = heapq.heappop(self.events)
@@ -155,6 +176,24 @@ class TraceReplay:
                 if job["pods"] or job["key"] not in self.jobs:
                     continue  # bound meanwhile or gone
                 self.try_schedule(job)
+            elif kind == "linkflap":
+                node, a, b = job
+                self.sim.alg.set_xgmi_link_healthy(node, a, b, False,
+                                                   self.rng.uniform(5.0, 40.0))
+                self.stats["link_degradations"] += 1
+            elif kind == "linkheal":
+                node, a, b = job
+                self.sim.alg.set_xgmi_link_healthy(node, a, b, True,
+                                                   self.rng.uniform(140.0, 155.0))
+                self.stats["link_heals"] += 1
+            elif kind == "gpufail":
+                node, g = job
+                self.sim.alg.set_leaf_cell_healthy(node, g, False)
+                self.stats["gpu_failures"] += 1
+            elif kind == "gpuheal":
+                node, g = job
+                self.sim.alg.set_leaf_cell_healthy(node, g, True)
+                self.stats["gpu_recoveries"] += 1
             elif kind == "finish":
                 if not job["pods"]:
                     continue  # was victimized; a retry event exists
@@ -169,6 +208,16 @@ class TraceReplay:
                 check()
                 self.stats["invariant_checks"] += 1
         # end-of-trace drain + VC-safety: every VC can take its full quota
+        # (heal any hardware weather still active first — safety is judged
+        # on a healthy cluster, as in the reference's semantics)
+        for i in range(self.nodes):
+            node = f"node{i + 1}"
+            self.sim.alg.set_healthy_node(node)
+            for g in range(8):
+                self.sim.alg.set_leaf_cell_healthy(node, g, True)
+            for l in self.sim.alg.get_xgmi_links(node):
+                if not l["healthy"]:
+                    self.sim.alg.set_xgmi_link_healthy(node, l["a"], l["b"], True, 153.0)
         for k in list(self.sim.pods):
             self.sim.delete_pod(k)
         violations = 0
