@@ -588,9 +588,6 @@ PhysicalCell* findPhysicalLeafCell(std::map<std::string, ChainCellList>& fullCel
                                    const std::string& chain, const std::string& node,
                                    int leafIndex);
 Cell* ancestorNoHigherThanNode(Cell* c);
-// max link-clean FREE leaf capacity within one cell (no session state);
-// suggestedNodes non-null restricts usable leaves to those nodes
-int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes = nullptr);
 // the chain's clean-shape worlds: one per enumerated consistent endpoint
 // choice (different max independent sets admit different clean shapes)
 std::vector<CleanShapeWorld> computeCleanShapeWorlds(

@@ -1,9 +1,11 @@
 // Standalone scheduling-core benchmark/profiling driver (no Python).
 // Builds an N-node MI355X cluster spec, replays a mixed 1/2/4-GPU guaranteed
 // workload with churn, and reports per-decision latency percentiles.
-// Compile (profiled):
-//   g++ -O2 -pg -std=c++17 -DHIVED_SCHED_BENCH_MAIN core/*.cpp? -> see Makefile note
-// Used to chase the verdict-8 target: 128-node filter p50 < 20 us.
+// Compile:
+//   g++ -O2 [-pg] -std=c++17 -o schedbench core/sched_bench_main.cpp \
+//       core/{cells,topo_sched,build,alloc,algorithm,debug}.cpp
+// Used to chase (and hold) the round-2 target: 128-node filter p50 < 20 us;
+// results in profiles/sched_scaling_r02.md.
 #include <algorithm>
 #include <chrono>
 #include <cstdio>

@@ -608,15 +608,6 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 
 }  // namespace
 
-// Max link-clean FREE leaf capacity within one cell: free healthy leaves
-// minus the minimum exclusions needed so no degraded link has both
-// endpoints in the set.
-int cleanFreeLeafCapacity(Cell* c, const std::set<std::string>* suggestedNodes) {
-  PickSession empty;
-  empty.suggestedFilter = suggestedNodes;
-  return cleanAvailAnalysis(c, kOpportunisticPriority, empty, INT_MAX, nullptr);
-}
-
 std::vector<CleanShapeWorld> computeCleanShapeWorlds(
     const ChainCellList& ccl, const std::set<std::string>* suggestedNodes, size_t maxWorlds) {
   PickSession base;
