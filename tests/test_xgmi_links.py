@@ -412,3 +412,36 @@ def test_property_multi_node_dirty_only_when_forced(seed):
                 if not any({a, b} <= set(ss) for a, b in links[n2]):
                     raise AssertionError(
                         f"dirty {sorted(cells)} on {node} but clean {ss} on {n2}")
+
+
+def test_pinned_cell_respects_degraded_link():
+    """Link avoidance inside a PINNED cell: a 2-GPU pinned request avoids
+    the degraded pair of its pinned node while the full 8-GPU pinned gang
+    still binds (capacity over quality)."""
+    from hivedscheduler_amd.api.types import PinnedCellSpec
+
+    cfg = mi355x_cluster_config(num_nodes=2, vcs={"VC1": [("MI355X-NODE", 1)]})
+    cfg.physicalCluster.physicalCells[1].pinnedCellId = "VC1-PIN"
+    cfg.virtualClusters["VC1"].pinnedCells = [PinnedCellSpec(pinnedCellId="VC1-PIN")]
+    sim = SimScheduler(cfg)
+    _mark_link(sim, "node2", 0, 1, False)
+    r = sim.schedule("ns/pin2", sim.pod_spec(leaf_cells=2, pinned_cell_id="VC1-PIN"))
+    assert r.kind == "bind" and r.bind_info.node == "node2"
+    assert sorted(r.bind_info.leafCellIsolation) != [0, 1], r.bind_info.leafCellIsolation
+    sim.delete_pod("ns/pin2")
+    r8 = sim.schedule("ns/pin8", sim.pod_spec(leaf_cells=8, pinned_cell_id="VC1-PIN"))
+    assert r8.kind == "bind" and sorted(r8.bind_info.leafCellIsolation) == list(range(8))
+    sim.alg._core.check_invariants()
+
+
+def test_opportunistic_hbm_demand():
+    """hbmBytesPerCell applies on the opportunistic (physical-view) path."""
+    cfg = mi355x_cluster_config(num_nodes=1)
+    leaf0 = cfg.physicalCluster.physicalCells[0].cellChildren[0].cellChildren[0].cellChildren[0]
+    leaf0.hbmBytes = 200 * GB
+    sim = SimScheduler(cfg)
+    r = sim.schedule("ns/ot", sim.pod_spec(leaf_cells=4, priority=-1,
+                                           hbm_bytes_per_cell=288 * GB))
+    assert r.kind == "bind"
+    assert 0 not in r.bind_info.leafCellIsolation
+    sim.alg._core.check_invariants()
