@@ -307,15 +307,22 @@ int cleanAvailAnalysis(Cell* viewCell, int p, const PickSession& s, int need,
   return clean;
 }
 
-// All maximum independent sets of the graph (masks), deduped, capped.
+// All maximum independent sets of the graph (masks), deduped by the
+// caller, branch-and-bound pruned: a branch that cannot reach the best
+// size seen so far is cut, so the maximum size is always found and only
+// maximum-size sets accumulate (include-first order reaches a maximal
+// set immediately, seeding the bound).
 void enumerateMaxIndepMasks(const std::vector<unsigned long long>& adj,
                             unsigned long long cand, unsigned long long acc,
                             std::vector<unsigned long long>& out, size_t cap) {
+  int best = 0;
+  for (auto m : out) best = std::max(best, __builtin_popcountll(m));
+  if (__builtin_popcountll(acc) + __builtin_popcountll(cand) < best) return;
   if (cand == 0) {
     out.push_back(acc);
     return;
   }
-  if (out.size() >= cap * 8) return;  // bound the search
+  if (out.size() >= cap * 32) return;  // hard stop on pathological graphs
   int v = __builtin_ctzll(cand);
   unsigned long long bit = 1ull << v;
   enumerateMaxIndepMasks(adj, cand & ~bit & ~adj[v], acc | bit, out, cap);
