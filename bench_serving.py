@@ -4,7 +4,7 @@ HTTP load on a live MI355X node.
 
 Runs uvicorn with the full scheduler, then drives it with concurrent HTTP
 clients executing complete pod lifecycles (filter -> bind -> delete) across
-three VCs with mixed sizes/priorities, while:
+two VCs with mixed sizes/priorities, while:
   - a health-agent loop runs REAL HIP kernels (bf16+fp8+fp4 MFMA, HBM triad,
     LDS) every ~30 s and posts reports;
   - xGMI link flaps and heals are injected through the health API;
@@ -19,7 +19,6 @@ import argparse
 import json
 import random
 import socket
-import statistics
 import threading
 import time
 
