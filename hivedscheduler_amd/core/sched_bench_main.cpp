@@ -62,9 +62,19 @@ static ClusterSpec makeSpec(int nodes) {
 int main(int argc, char** argv) {
   int nodes = argc > 1 ? atoi(argv[1]) : 128;
   int nreq = argc > 2 ? atoi(argv[2]) : 2000;
+  int badLinks = argc > 3 ? atoi(argv[3]) : 0;  // degraded links to scatter
   auto tb0 = std::chrono::steady_clock::now();
   HivedCore core(makeSpec(nodes));
   for (int i = 0; i < nodes; i++) core.setNodeHealthy("node" + std::to_string(i + 1), true);
+  {
+    std::mt19937 lrng(7);
+    for (int i = 0; i < badLinks; i++) {
+      int n = (int)(lrng() % nodes) + 1;
+      int a = (int)(lrng() % 8), b = (int)(lrng() % 8);
+      if (a == b) b = (a + 1) % 8;
+      core.setXgmiLinkHealthy("node" + std::to_string(n), a, b, false, 12.0);
+    }
+  }
   auto tb1 = std::chrono::steady_clock::now();
 
   std::mt19937 rng(0);
@@ -102,7 +112,8 @@ int main(int argc, char** argv) {
     }
   }
   std::sort(lat.begin(), lat.end());
-  printf("nodes=%d reqs=%d build=%.1fms p50=%.1fus p90=%.1fus p99=%.1fus\n", nodes, nreq,
+  printf("nodes=%d reqs=%d badLinks=%d build=%.1fms p50=%.1fus p90=%.1fus p99=%.1fus\n", nodes, nreq,
+         badLinks,
          std::chrono::duration<double, std::milli>(tb1 - tb0).count(), lat[lat.size() / 2],
          lat[(size_t)(lat.size() * 0.90)], lat[(size_t)(lat.size() * 0.99)]);
   return 0;

@@ -628,6 +628,25 @@ std::vector<CleanShapeWorld> computeCleanShapeWorlds(
     if (!variants.empty()) perCell.push_back(std::move(variants));
   }
   std::vector<CleanShapeWorld> worlds;
+  // shared across worlds: the unbound-cell lists, and each clean cell's
+  // capacity via the O(1) counter fast path (exclusions only ever sit
+  // under bad-link cells, so cells with badLinksUnder == 0 are unaffected
+  // by any world's exclusion set — this keeps the per-schedule world
+  // computation at O(bad-link subtrees), not O(chain), per world)
+  std::map<int, std::vector<PhysicalCell*>> physByLevel;
+  std::map<int, int> cleanCellBest;  // per level, max capacity among clean cells
+  PickSession noExcl = base;
+  for (int l = kLowestLevel; l <= top; l++) {
+    int best = 0;
+    for (Cell* c : ccl.at(l)) {
+      auto* ph = static_cast<PhysicalCell*>(c);
+      if (ph->virt == nullptr) physByLevel[l].push_back(ph);
+      if (ph->badLinksUnder == 0) {
+        best = std::max(best, availLeaves(c, kOpportunisticPriority, noExcl).first);
+      }
+    }
+    cleanCellBest[l] = best;
+  }
   // odometer over the per-cell variants, capped
   std::vector<size_t> idx(perCell.size(), 0);
   for (;;) {
@@ -640,18 +659,16 @@ std::vector<CleanShapeWorld> computeCleanShapeWorlds(
       }
     }
     for (int l = kLowestLevel; l <= top; l++) {
-      int best = 0;
+      int best = cleanCellBest[l];
       for (Cell* c : ccl.at(l)) {
-        best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+        auto* ph = static_cast<PhysicalCell*>(c);
+        if (ph->badLinksUnder > 0) {
+          best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+        }
       }
       w.caps[l] = best;
     }
-    for (int l = kLowestLevel; l <= top; l++) {
-      for (Cell* c : ccl.at(l)) {
-        auto* ph = static_cast<PhysicalCell*>(c);
-        if (ph->virt == nullptr) w.physByLevel[l].push_back(ph);
-      }
-    }
+    w.physByLevel = physByLevel;
     worlds.push_back(std::move(w));
     if (worlds.size() >= maxWorlds || perCell.empty()) break;
     // advance the odometer
