@@ -69,6 +69,14 @@ struct PickSession {
   // restrict leaf usability to these nodes (caps computation honors the
   // K8s suggested-node set the mapping stage will enforce)
   const std::set<std::string>* suggestedFilter = nullptr;
+  // physical leaves consumed by earlier world-mode picks of THIS session
+  // (gang pods share the session): the hint math must not offer the same
+  // physical capacity to two pods — static in-world counts made pod 3's
+  // matching hint a pair whose leaves pod 2 had already spoken for, and
+  // the mapping then had no clean assignment (found by the gang oracle)
+  std::unordered_set<PhysicalCell*> physTaken;
+  // virtual leaf -> the hint leaf it consumed (for exact rollback)
+  std::unordered_map<Cell*, PhysicalCell*> hintTaken;
   // link-honoring attempts on VIRTUAL views only: the chain's clean-shape
   // world. Excluded physical leaves are unavailable (applied to bound
   // leaves directly and to bound regions via a physical min-cap); unbound
@@ -80,24 +88,28 @@ struct PickSession {
   const CleanShapeWorld* world = nullptr;
 };
 
-// in-world free capacity of a PHYSICAL region (free + healthy + not excluded)
-int physFreeInWorld(PhysicalCell* pc, const CleanShapeWorld& w) {
+// in-world free capacity of a PHYSICAL region (free + healthy + not
+// excluded + not consumed by an earlier world-mode pick of this session)
+int physFreeInWorld(PhysicalCell* pc, const PickSession& s) {
   if (pc->level == kLowestLevel) {
-    return (pc->priority == kFreePriority && pc->healthy && !w.excluded.count(pc)) ? 1 : 0;
+    return (pc->priority == kFreePriority && pc->healthy && !s.world->excluded.count(pc) &&
+            (s.physTaken.empty() || !s.physTaken.count(pc)))
+               ? 1
+               : 0;
   }
   int n = 0;
-  for (Cell* child : pc->children) n += physFreeInWorld(static_cast<PhysicalCell*>(child), w);
+  for (Cell* child : pc->children) n += physFreeInWorld(static_cast<PhysicalCell*>(child), s);
   return n;
 }
 
 // best in-world capacity of any single UNBOUND physical cell at `level`
 // under region `pc` (an unbound virtual cell whose nearest bound ancestor
 // is bound to pc will map onto one of these)
-int maxInWorldAtLevelUnder(PhysicalCell* pc, int level, const CleanShapeWorld& w) {
-  if (pc->level == level) return pc->virt == nullptr ? physFreeInWorld(pc, w) : 0;
+int maxInWorldAtLevelUnder(PhysicalCell* pc, int level, const PickSession& s) {
+  if (pc->level == level) return pc->virt == nullptr ? physFreeInWorld(pc, s) : 0;
   int best = 0;
   for (Cell* child : pc->children) {
-    best = std::max(best, maxInWorldAtLevelUnder(static_cast<PhysicalCell*>(child), level, w));
+    best = std::max(best, maxInWorldAtLevelUnder(static_cast<PhysicalCell*>(child), level, s));
   }
   return best;
 }
@@ -187,7 +199,7 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
       }
       int cap = INT_MAX;
       if (region != nullptr) {
-        cap = maxInWorldAtLevelUnder(region, c->level, *s.world);
+        cap = maxInWorldAtLevelUnder(region, c->level, s);
       } else {
         auto it = s.world->caps.find(c->level);
         if (it != s.world->caps.end()) cap = it->second;
@@ -200,7 +212,7 @@ std::pair<int, int> availLeaves(Cell* c, int p, const PickSession& s) {
       // bound: the physical region is known; its in-world free capacity
       // bounds what the subtree (incl. positionally-ambiguous unbound
       // leaves) can deliver cleanly
-      int cap = physFreeInWorld(ph, *s.world);
+      int cap = physFreeInWorld(ph, s);
       // preemptible leaves are outside the free-world analysis; only the
       // free component is capped, the preemptible surplus rides on top
       int preemptible = at - af;
@@ -472,23 +484,19 @@ void pickLeaves(Cell* cell, int q, int p, PickSession& s, std::vector<Cell*>& ou
 // bindings are upward-contiguous) are rank-matched to the hint's unbound
 // children by capacity. Returns false (instead of throwing) when the shape
 // cannot be realized — the caller falls to the next world / dirty rung.
-bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& s,
-                     std::vector<Cell*>& out) {
-  if (cell->level == kLowestLevel) {
-    // re-check eligibility (the caller's counts came from caps, not hints)
-    auto [t, f] = availLeaves(cell, p, s);
-    (void)f;
-    if (t < 1) return false;
-    s.taken.insert(cell);
-    out.push_back(cell);
-    return true;
-  }
-  struct Pair {
-    Cell* child;
-    PhysicalCell* hint;
-    int avail;
-  };
-  std::vector<Pair> pairs;
+// Children of a world-mode (cell, hint) pair, matched: bound virtual
+// children identify their hint child directly; unbound ones (all-unbound
+// subtrees, since bindings are upward-contiguous) are rank-matched to the
+// hint's unbound children by capacity. Used by BOTH the availability
+// recursion and the descent so their answers agree.
+struct WorldPair {
+  Cell* child;
+  PhysicalCell* hint;
+  int avail;
+};
+
+bool matchWorldChildren(Cell* cell, PhysicalCell* hint, int p, const PickSession& s,
+                        std::vector<WorldPair>& pairs) {
   std::vector<Cell*> unboundChildren;
   std::vector<PhysicalCell*> freeHints;
   for (Cell* childC : cell->children) {
@@ -504,12 +512,8 @@ bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& 
     if (ph->virt == nullptr) freeHints.push_back(ph);
   }
   if (freeHints.size() < unboundChildren.size()) return false;
-  // rank-match: unbound virtual children are interchangeable, so order by
-  // the hint children's in-world capacity descending (tightest structure
-  // last). Sort virtual children by raw availability desc to pair the most
-  // capable with the most capable.
   std::stable_sort(freeHints.begin(), freeHints.end(), [&](PhysicalCell* a, PhysicalCell* b) {
-    return physFreeInWorld(a, *s.world) > physFreeInWorld(b, *s.world);
+    return physFreeInWorld(a, s) > physFreeInWorld(b, s);
   });
   std::stable_sort(unboundChildren.begin(), unboundChildren.end(), [&](Cell* a, Cell* b) {
     return availLeaves(a, p, s).first > availLeaves(b, p, s).first;
@@ -517,9 +521,46 @@ bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& 
   for (size_t i = 0; i < unboundChildren.size(); i++) {
     pairs.push_back({unboundChildren[i], freeHints[i], 0});
   }
+  return true;
+}
+
+// RECURSIVE world availability of a matched (cell, hint) pair: the virtual
+// and physical structures are intersected level by level. A parent-level
+// min(virtualAvail, physInWorld) is NOT enough — with session-taken leaves
+// from earlier gang pods, the virtual side's remaining leaves and the
+// hint's in-world leaves can sit in different sub-cells, and only the
+// recursive intersection sees it (found by the gang-union oracle: a
+// 3x2-pod gang claimed a quad could serve pod 2 when its pairs could
+// deliver only 1 leaf each).
+int availWorld(Cell* cell, PhysicalCell* hint, int p, const PickSession& s) {
+  if (cell->level == kLowestLevel) {
+    auto [t, f] = availLeaves(cell, p, s);
+    (void)f;
+    return std::min(t, physFreeInWorld(hint, s));
+  }
+  std::vector<WorldPair> pairs;
+  if (!matchWorldChildren(cell, hint, p, s, pairs)) return 0;
+  int total = 0;
+  for (auto& pr : pairs) total += availWorld(pr.child, pr.hint, p, s);
+  return total;
+}
+
+bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& s,
+                     std::vector<Cell*>& out) {
+  if (cell->level == kLowestLevel) {
+    // re-check eligibility (the caller's counts came from caps, not hints)
+    if (availWorld(cell, hint, p, s) < 1) return false;
+    s.taken.insert(cell);
+    s.physTaken.insert(hint);
+    s.hintTaken[cell] = hint;
+    out.push_back(cell);
+    return true;
+  }
+  std::vector<WorldPair> pairs;
+  if (!matchWorldChildren(cell, hint, p, s, pairs)) return false;
   int total = 0;
   for (auto& pr : pairs) {
-    pr.avail = std::min(availLeaves(pr.child, p, s).first, physFreeInWorld(pr.hint, *s.world));
+    pr.avail = availWorld(pr.child, pr.hint, p, s);
     total += pr.avail;
   }
   if (total < q) return false;
@@ -535,7 +576,7 @@ bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& 
     return pickLeavesWorld(pairs[best].child, pairs[best].hint, q, p, s, out);
   }
   // drain, largest first
-  std::stable_sort(pairs.begin(), pairs.end(), [](const Pair& a, const Pair& b) {
+  std::stable_sort(pairs.begin(), pairs.end(), [](const WorldPair& a, const WorldPair& b) {
     return a.avail > b.avail;
   });
   int remaining = q;
@@ -545,8 +586,15 @@ bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& 
     int t = std::min(remaining, pr.avail);
     if (t > 0) {
       if (!pickLeavesWorld(pr.child, pr.hint, t, p, s, out)) {
-        // roll back this request's takes and report shape failure
-        for (size_t i = outStart; i < out.size(); i++) s.taken.erase(out[i]);
+        // roll back this request's takes (virtual AND hint-leaf claims)
+        for (size_t i = outStart; i < out.size(); i++) {
+          s.taken.erase(out[i]);
+          auto it = s.hintTaken.find(out[i]);
+          if (it != s.hintTaken.end()) {
+            s.physTaken.erase(it->second);
+            s.hintTaken.erase(it);
+          }
+        }
         out.resize(outStart);
         return false;
       }
@@ -554,20 +602,29 @@ bool pickLeavesWorld(Cell* cell, PhysicalCell* hint, int q, int p, PickSession& 
     }
   }
   if (remaining != 0) {
-    for (size_t i = outStart; i < out.size(); i++) s.taken.erase(out[i]);
+    for (size_t i = outStart; i < out.size(); i++) {
+      s.taken.erase(out[i]);
+      auto it = s.hintTaken.find(out[i]);
+      if (it != s.hintTaken.end()) {
+        s.physTaken.erase(it->second);
+        s.hintTaken.erase(it);
+      }
+    }
     out.resize(outStart);
     return false;
   }
   return true;
 }
 
-// Resolve the tentative physical target for a world-mode pick at `cell`:
-// its own binding if bound; otherwise the tightest-fit unbound physical
-// cell with in-world capacity >= q — scoped under the nearest bound
+// Candidate physical targets for a world-mode pick at `cell`: its own
+// binding if bound; otherwise the unbound physical cells with static
+// in-world capacity >= q, tightest first — scoped under the nearest bound
 // ancestor's region, or chain-wide via the world's per-level cell list.
-PhysicalCell* resolveWorldHint(Cell* cell, int q, const PickSession& s) {
+// The caller tries them in order: the static capacity is an upper bound
+// and the recursive availWorld intersection can reject a candidate.
+std::vector<PhysicalCell*> resolveWorldHints(Cell* cell, int q, const PickSession& s) {
   auto* vc = static_cast<VirtualCell*>(cell);
-  if (vc->phys != nullptr) return vc->phys;
+  if (vc->phys != nullptr) return {vc->phys};
   PhysicalCell* region = nullptr;
   for (Cell* a = cell->parent; a != nullptr; a = a->parent) {
     PhysicalCell* ap = static_cast<VirtualCell*>(a)->phys;
@@ -576,15 +633,10 @@ PhysicalCell* resolveWorldHint(Cell* cell, int q, const PickSession& s) {
       break;
     }
   }
-  PhysicalCell* best = nullptr;
-  int bestCap = INT_MAX;
+  std::vector<PhysicalCell*> candidates;
   auto consider = [&](PhysicalCell* ph) {
     if (ph->virt != nullptr) return;
-    int cap = physFreeInWorld(ph, *s.world);
-    if (cap >= q && cap < bestCap) {
-      best = ph;
-      bestCap = cap;
-    }
+    if (physFreeInWorld(ph, s) >= q) candidates.push_back(ph);
   };
   if (region != nullptr) {
     std::function<void(PhysicalCell*)> walk = [&](PhysicalCell* ph) {
@@ -601,7 +653,11 @@ PhysicalCell* resolveWorldHint(Cell* cell, int q, const PickSession& s) {
       for (PhysicalCell* ph : it->second) consider(ph);
     }
   }
-  return best;
+  std::stable_sort(candidates.begin(), candidates.end(),
+                   [&](PhysicalCell* a, PhysicalCell* b) {
+                     return physFreeInWorld(a, s) < physFreeInWorld(b, s);
+                   });
+  return candidates;
 }
 
 std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>& suggestedNodes,
@@ -810,10 +866,17 @@ bool TopoScheduler::tryScheduleAtPriority(const std::vector<int>& sortedLeafNums
     std::vector<Cell*> leaves;
     leaves.reserve(q);
     if (session.world != nullptr && !node->isPhysical()) {
-      // world mode on a virtual view: physical-mirroring descent; a shape
-      // that cannot be realized fails the attempt (next world / dirty rung)
-      PhysicalCell* hint = resolveWorldHint(node, q, session);
-      if (hint == nullptr || !pickLeavesWorld(node, hint, q, priority, session, leaves)) {
+      // world mode on a virtual view: physical-mirroring descent over
+      // tightest-first candidate hints; a shape no candidate can realize
+      // fails the attempt (next world / dirty rung)
+      bool placed = false;
+      for (PhysicalCell* hint : resolveWorldHints(node, q, session)) {
+        if (pickLeavesWorld(node, hint, q, priority, session, leaves)) {
+          placed = true;
+          break;
+        }
+      }
+      if (!placed) {
         *failedReason = "no link-clean shape in this world";
         return false;
       }

@@ -445,3 +445,45 @@ def test_opportunistic_hbm_demand():
     assert r.kind == "bind"
     assert 0 not in r.bind_info.leafCellIsolation
     sim.alg._core.check_invariants()
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_property_gang_union_dirty_only_when_forced(seed):
+    """Gang oracle: ALL pods of an affinity group communicate (the group's
+    collective), so the UNION of their placements must avoid degraded links
+    unless no clean union of that size exists — exact over 3000 random
+    multi-pod gang cases offline (found and fixed two session-state bugs:
+    non-recursive hint availability and cross-pod hint collisions)."""
+    import itertools
+    import random
+
+    rng = random.Random(300000 + seed)
+    sim = SimScheduler(mi355x_cluster_config(num_nodes=1))
+    for i in range(rng.randrange(0, 3)):
+        sim.schedule(f"occ/p{i}", sim.pod_spec(leaf_cells=1))
+    used = {i for k, (sp, info) in sim.pods.items() for i in info.leafCellIsolation}
+    free = set(range(8)) - used
+    links = set()
+    for _ in range(rng.randrange(1, 4)):
+        a, b = rng.sample(range(8), 2)
+        links.add((min(a, b), max(a, b)))
+    for a, b in links:
+        sim.alg.set_xgmi_link_healthy("node1", a, b, False)
+    pods, size = rng.choice([(2, 2), (2, 3), (3, 2), (2, 4), (4, 2)])
+    spec = sim.pod_spec(leaf_cells=size, group="g", members=[(pods, size)])
+    cells = []
+    for i in range(pods):
+        r = sim.schedule(f"g/p{i}", spec)
+        if r.kind != "bind":
+            sim.alg._core.check_invariants()
+            return
+        cells.extend(r.bind_info.leafCellIsolation)
+    sim.alg._core.check_invariants()
+    cset = set(cells)
+    if any({a, b} <= cset for a, b in links):
+        q = pods * size
+        for ss in itertools.combinations(sorted(free), q):
+            if not any({a, b} <= set(ss) for a, b in links):
+                raise AssertionError(
+                    f"gang union {sorted(cells)} dirty but clean {ss} existed "
+                    f"(links {sorted(links)})")
