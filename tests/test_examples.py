@@ -42,10 +42,14 @@ def test_example_request_pods_parse_and_validate():
     files = sorted(glob.glob(os.path.join(REPO, "examples/request/*.yaml")))
     assert len(files) >= 4
     for f in files:
-        p = yaml.safe_load(open(f))
-        assert podmod.is_hived_enabled(p), f
-        spec = podmod.extract_pod_scheduling_spec(p)
-        assert spec.virtualCluster and spec.leafCellNumber > 0, f
+        # multi-document files (e.g. the distributed-training gang) contain
+        # several pods; every one must validate
+        for p in yaml.safe_load_all(open(f)):
+            assert podmod.is_hived_enabled(p), f
+            spec = podmod.extract_pod_scheduling_spec(p)
+            assert spec.virtualCluster and spec.leafCellNumber > 0, f
+            if "gang" in f:
+                assert spec.affinityGroup and spec.affinityGroup.members[0].podNumber == 2
         if "legacy" in f:
             assert spec.leafCellType == "MI355X" and spec.leafCellNumber == 4
         if "pinned" in f:
