@@ -534,16 +534,28 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
   if (chainHasBadLinks(sr.chain) && gangLeaves >= 2) {
     const std::set<std::string>* filter =
         sr.ignoreSuggestedNodes ? nullptr : sr.suggestedNodes;
-    auto worlds = computeCleanShapeWorlds(fullCellList_[sr.chain], filter, 4);
-    for (auto& w : worlds) {
-      sr.cleanWorld = &w;
-      if (mapDebugRelease()) {
-        fprintf(stderr, "[world] chain=%s excl=%zu:", sr.chain.c_str(), w.excluded.size());
-        for (auto& [l, v] : w.caps) fprintf(stderr, " L%d=%d", l, v);
-        fprintf(stderr, "\n");
+    // Link-clean placements take precedence over dirty ones even when
+    // cleanliness costs a preemption: a gang all-reducing over a degraded
+    // link (~12 vs ~153 GB/s) is crippled for its whole lifetime, while
+    // preemption (usually of opportunistic scavengers) is the scheduler's
+    // normal business. Tier order: all-free clean worlds, then preemptive
+    // clean worlds, then the unconstrained fallback ladder.
+    auto tryWorlds = [&](int tier) {
+      auto worlds = computeCleanShapeWorlds(fullCellList_[sr.chain], filter, 4, tier);
+      for (auto& w : worlds) {
+        sr.cleanWorld = &w;
+        if (mapDebugRelease()) {
+          fprintf(stderr, "[world] chain=%s tier=%d excl=%zu:", sr.chain.c_str(), tier,
+                  w.excluded.size());
+          for (auto& [l, v] : w.caps) fprintf(stderr, " L%d=%d", l, v);
+          fprintf(stderr, "\n");
+        }
+        if (attemptOnce(true)) return true;
       }
-      if (attemptOnce(true)) return true;
-    }
+      return false;
+    };
+    if (tryWorlds(kOpportunisticPriority)) return true;
+    if (sr.priority > kOpportunisticPriority && tryWorlds(sr.priority)) return true;
     sr.cleanWorld = nullptr;
   }
   if (attemptOnce(false)) return true;

@@ -280,6 +280,10 @@ using Placement = std::map<int, std::vector<std::vector<CellT*>>>;
 // the degraded-link graph restricted to free leaves); `caps` = per-level max
 // in-world free capacity of any single physical cell.
 struct CleanShapeWorld {
+  // the scheduling tier this world models: in-world availability counts
+  // leaves below this priority (kOpportunisticPriority = free cells only;
+  // a request's own priority = free + preemptible)
+  int priority = kOpportunisticPriority;
   std::map<int, int> caps;
   std::unordered_set<PhysicalCell*> excluded;
   // unbound physical cells per level, for hint resolution when no bound
@@ -592,7 +596,7 @@ Cell* ancestorNoHigherThanNode(Cell* c);
 // choice (different max independent sets admit different clean shapes)
 std::vector<CleanShapeWorld> computeCleanShapeWorlds(
     const ChainCellList& ccl, const std::set<std::string>* suggestedNodes,
-    size_t maxWorlds = 4);
+    size_t maxWorlds = 4, int priority = kOpportunisticPriority);
 void checkInvariants(const HivedCore& core);
 
 }  // namespace hived

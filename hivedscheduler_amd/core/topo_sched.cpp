@@ -92,7 +92,7 @@ struct PickSession {
 // excluded + not consumed by an earlier world-mode pick of this session)
 int physFreeInWorld(PhysicalCell* pc, const PickSession& s) {
   if (pc->level == kLowestLevel) {
-    return (pc->priority == kFreePriority && pc->healthy && !s.world->excluded.count(pc) &&
+    return (pc->priority < s.world->priority && pc->healthy && !s.world->excluded.count(pc) &&
             (s.physTaken.empty() || !s.physTaken.count(pc)))
                ? 1
                : 0;
@@ -346,15 +346,15 @@ void enumerateMaxIndepMasks(const std::vector<unsigned long long>& adj,
 // different choices admit different clean shapes, so the ladder tries a few.
 std::vector<std::vector<Cell*>> enumerateCleanExclusionVariants(Cell* top,
                                                                 const PickSession& s,
-                                                                size_t cap) {
+                                                                size_t cap, int priority) {
   std::vector<std::vector<Cell*>> out;
   if (badLinksUnderOf(top) == 0) return out;
-  BadLinkGraph g = buildBadLinkGraph(top, kOpportunisticPriority, s);
+  BadLinkGraph g = buildBadLinkGraph(top, priority, s);
   int nv = static_cast<int>(g.verts.size());
   if (nv == 0 || nv > 24) {
     if (nv > 24) {  // degenerate: one conservative variant via the analysis
       std::vector<Cell*> excl;
-      cleanAvailAnalysis(top, kOpportunisticPriority, s, 0, &excl);
+      cleanAvailAnalysis(top, priority, s, 0, &excl);
       if (!excl.empty()) out.push_back(std::move(excl));
     }
     return out;
@@ -672,7 +672,8 @@ std::tuple<bool, bool> healthyAndSuggested(Cell* c, const std::set<std::string>&
 }  // namespace
 
 std::vector<CleanShapeWorld> computeCleanShapeWorlds(
-    const ChainCellList& ccl, const std::set<std::string>* suggestedNodes, size_t maxWorlds) {
+    const ChainCellList& ccl, const std::set<std::string>* suggestedNodes, size_t maxWorlds,
+    int priority) {
   PickSession base;
   base.suggestedFilter = suggestedNodes;
   int top = ccl.top();
@@ -680,7 +681,7 @@ std::vector<CleanShapeWorld> computeCleanShapeWorlds(
   // its bad-link graph). Links are node-local, so top cells are independent.
   std::vector<std::vector<std::vector<Cell*>>> perCell;
   for (Cell* c : ccl.at(top)) {
-    auto variants = enumerateCleanExclusionVariants(c, base, maxWorlds);
+    auto variants = enumerateCleanExclusionVariants(c, base, maxWorlds, priority);
     if (!variants.empty()) perCell.push_back(std::move(variants));
   }
   std::vector<CleanShapeWorld> worlds;
@@ -698,7 +699,7 @@ std::vector<CleanShapeWorld> computeCleanShapeWorlds(
       auto* ph = static_cast<PhysicalCell*>(c);
       if (ph->virt == nullptr) physByLevel[l].push_back(ph);
       if (ph->badLinksUnder == 0) {
-        best = std::max(best, availLeaves(c, kOpportunisticPriority, noExcl).first);
+        best = std::max(best, availLeaves(c, priority, noExcl).first);
       }
     }
     cleanCellBest[l] = best;
@@ -714,12 +715,13 @@ std::vector<CleanShapeWorld> computeCleanShapeWorlds(
         w.excluded.insert(static_cast<PhysicalCell*>(e));
       }
     }
+    w.priority = priority;
     for (int l = kLowestLevel; l <= top; l++) {
       int best = cleanCellBest[l];
       for (Cell* c : ccl.at(l)) {
         auto* ph = static_cast<PhysicalCell*>(c);
         if (ph->badLinksUnder > 0) {
-          best = std::max(best, availLeaves(c, kOpportunisticPriority, session).first);
+          best = std::max(best, availLeaves(c, priority, session).first);
         }
       }
       w.caps[l] = best;
@@ -918,11 +920,12 @@ bool TopoScheduler::Schedule(const std::map<int, int>& podLeafCellNums, int prio
     return tryScheduleAtPriority(sortedLeafNums, p, suggestedNodes, ignoreSuggestedNodes,
                                  minHbmBytes, honorLinks, cleanWorld, out, failedReason);
   };
+  // world-mode (honorOnly) attempts run ONLY at the world's own tier: the
+  // caller iterates worlds built at kOpportunistic (free cells) and, if
+  // those fail, worlds built at the request's priority (free +
+  // preemptible); each world's availability math matches its tier.
+  if (honorOnly) return attempt(cleanWorld != nullptr ? cleanWorld->priority : priority, true);
   if (attempt(kOpportunisticPriority, true)) return true;
-  // world-mode (honorOnly) attempts run at the FREE tier only: the world is
-  // built from free leaves and its hint math does not model preemptible
-  // cells; preemptive clean placement falls to the no-world honor rung
-  if (honorOnly) return false;
   if (anyBadLinks && attempt(kOpportunisticPriority, false)) return true;
   if (priority > kOpportunisticPriority) {
     if (attempt(priority, true)) return true;
