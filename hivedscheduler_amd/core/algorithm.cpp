@@ -541,8 +541,24 @@ bool HivedCore::scheduleGuaranteedGroup(const SchedulingRequest& srIn, Placement
     // normal business. Tier order: all-free clean worlds, then preemptive
     // clean worlds, then the unconstrained fallback ladder.
     auto tryWorlds = [&](int tier) {
-      auto worlds = computeCleanShapeWorlds(fullCellList_[sr.chain], filter, 4, tier);
-      for (auto& w : worlds) {
+      // cache worlds per (chain, tier) while no world-relevant mutation
+      // happened; failed filters of waiting pods then skip the rebuild.
+      // Requests with a suggested-node restriction bypass the cache (the
+      // filter changes the world).
+      std::vector<CleanShapeWorld> computed;
+      const std::vector<CleanShapeWorld>* worldsPtr;
+      if (filter == nullptr) {
+        WorldCacheEntry& entry = worldCache_[{sr.chain, tier}];
+        if (entry.epoch != gWorldEpochCounter) {
+          entry.worlds = computeCleanShapeWorlds(fullCellList_[sr.chain], nullptr, 4, tier);
+          entry.epoch = gWorldEpochCounter;
+        }
+        worldsPtr = &entry.worlds;
+      } else {
+        computed = computeCleanShapeWorlds(fullCellList_[sr.chain], filter, 4, tier);
+        worldsPtr = &computed;
+      }
+      for (const auto& w : *worldsPtr) {
         sr.cleanWorld = &w;
         if (mapDebugRelease()) {
           fprintf(stderr, "[world] chain=%s tier=%d excl=%zu:", sr.chain.c_str(), tier,

@@ -677,6 +677,7 @@ void HivedCore::setXgmiLinkHealthy(const std::string& node, int a, int b, bool h
   if (gbps > 0) rec.gbps = gbps;
   rec.healthy = healthy;
   if (healthy == wasHealthy) return;
+  gWorldEpochCounter++;
   if (healthy) {
     auto erasePeer = [](PhysicalCell* from, PhysicalCell* peer) {
       auto& v = from->badLinkPeers;
@@ -713,6 +714,7 @@ std::vector<std::tuple<int, int, double, bool>> HivedCore::xgmiLinks(
 // A cell is bad if ANY child is bad; propagate from leaf up.
 void HivedCore::setBadCell(PhysicalCell* c) {
   if (!c->healthy) return;
+  gWorldEpochCounter++;
   c->healthy = false;
   if (c->parent != nullptr) setBadCell(static_cast<PhysicalCell*>(c->parent));
   if (inFreeCellList(c)) {
@@ -737,6 +739,7 @@ void HivedCore::setBadCell(PhysicalCell* c) {
 // A cell is healthy if ALL children are healthy; propagate from leaf up.
 void HivedCore::setHealthyCell(PhysicalCell* c) {
   if (c->healthy) return;
+  gWorldEpochCounter++;
   c->healthy = true;
   if (inFreeCellList(c)) {
     removeBadFreeCell(c);

@@ -24,10 +24,19 @@ const char* to_string(GState s) {
   return "?";
 }
 
+// Global mutation counter for the clean-shape-world cache: bumped by every
+// mutation that can change a world (leaf priorities, bindings, health,
+// links). Process-global on purpose — with several cores alive (tests) a
+// foreign bump merely costs a cache miss, never staleness.
+unsigned long long gWorldEpochCounter = 1;
+
+static inline void bumpWorldEpoch() { gWorldEpochCounter++; }
+
 // Priority of a cell is the max of its children's; propagate from leaf up.
 void setCellPriority(Cell* c, int p) {
   int original = c->priority;
   c->priority = p;
+  if (c->level == kLowestLevel && original != p) bumpWorldEpoch();
   // maintain the freeLeavesUnder cache on leaf free<->used transitions
   if (c->level == kLowestLevel && (original == kFreePriority) != (p == kFreePriority)) {
     int delta = p == kFreePriority ? 1 : -1;
@@ -59,6 +68,7 @@ void updateUsedLeafCellNumAtPriority(Cell* c, int p, bool increase) {
 // partner with a dangling phys reference (fuzz-found asymmetric binding
 // after a lazy-preemption revert whose victim kept bad-cell bindings).
 void bindCell(PhysicalCell* pc, VirtualCell* vc) {
+  bumpWorldEpoch();
   while (vc->phys == nullptr && pc->virt == nullptr) {
     pc->virt = vc;
     vc->phys = pc;
@@ -80,6 +90,7 @@ void bindCell(PhysicalCell* pc, VirtualCell* vc) {
 // left the binding and the free list out of sync, and a later re-bind
 // double-allocated the cell.
 void unbindCell(PhysicalCell* c) {
+  bumpWorldEpoch();
   VirtualCell* boundVirtual = c->virt;
   while (boundVirtual->phys != nullptr && !boundVirtual->phys->pinned) {
     PhysicalCell* boundPhysical = boundVirtual->phys;

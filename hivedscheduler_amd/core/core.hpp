@@ -575,9 +575,20 @@ class HivedCore {
   // node name -> leaf cells on that node (for health propagation)
   std::map<std::string, std::vector<PhysicalCell*>> nodeLeafCellsStorage_;
   long long scheduleCount_ = 0;
+  // Clean-shape-world cache, validated against gWorldEpochCounter (bumped
+  // by every mutation that can change a world: leaf priorities, bindings,
+  // health, links). Wait-storms — the common case under contention: many
+  // failed filters with no allocation in between — reuse cached worlds
+  // instead of recomputing them per Schedule.
+  struct WorldCacheEntry {
+    unsigned long long epoch = 0;
+    std::vector<CleanShapeWorld> worlds;
+  };
+  std::map<std::pair<std::string, int>, WorldCacheEntry> worldCache_;
 };
 
 // helpers shared across translation units
+extern unsigned long long gWorldEpochCounter;
 void setCellPriority(Cell* c, int p);
 void updateUsedLeafCellNumAtPriority(Cell* c, int p, bool increase);
 void bindCell(PhysicalCell* pc, VirtualCell* vc);
